@@ -1,0 +1,53 @@
+/* SPDX-License-Identifier: GPL-2.0 OR MIT
+ *
+ * fake_kfd.h — userspace stand-in for the amdkfd amd_rdma interface
+ * (contract: module/include/rocnr_amd_rdma.h).  Lets the bridge/probe
+ * modules run real pin/unpin/invalidate flows, including the async revoke
+ * race, with no GPU.
+ *
+ * Contract implemented (documented in the reference at
+ * /root/reference/amdp2p.c:105-107 and re-stated here):
+ *  - get_pages pins [va, va+size) of a fake GPU allocation and returns an
+ *    amd_p2p_info whose sg table holds synthetic bus addresses at 2 MiB
+ *    granularity (optionally fragmented, to exercise coalescing);
+ *  - fake_kfd_free() revokes an allocation: each live pin's free_callback
+ *    is invoked, then the pin's resources are reclaimed;
+ *  - put_pages serializes against in-flight free callbacks for the same
+ *    pin (waits for them), as a real KFD must for the peer contract to be
+ *    sound.
+ */
+#ifndef ROCNR_FAKE_KFD_H_
+#define ROCNR_FAKE_KFD_H_
+
+#include "rocnr_shim_all.h"
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+#define FAKE_KFD_VRAM_PAGE (2ULL << 20)	/* MI355X VRAM granule */
+
+/* Create a fake GPU allocation.  frag_every > 0 inserts a bus-address
+ * hole after every frag_every-th 2 MiB chunk (so a pin coalesces into
+ * ceil(nchunks / frag_every) runs); 0 = fully bus-contiguous.  Returns
+ * the GPU VA. */
+uint64_t fake_kfd_alloc(uint64_t size, unsigned int frag_every);
+
+/* Free an allocation: fires free callbacks on all live pins (the async
+ * invalidation path), then reclaims them.  Safe from any thread. */
+void fake_kfd_free(uint64_t va);
+
+void fake_kfd_reset(void);
+
+/* Introspection for tests. */
+long fake_kfd_live_pins(void);
+long fake_kfd_get_pages_calls(void);
+long fake_kfd_put_pages_calls(void);
+long fake_kfd_bad_put_calls(void);	/* put of unknown pin = bridge bug */
+long fake_kfd_callbacks_fired(void);
+
+#ifdef __cplusplus
+}
+#endif
+
+#endif /* ROCNR_FAKE_KFD_H_ */

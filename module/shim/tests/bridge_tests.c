@@ -1,0 +1,382 @@
+/* SPDX-License-Identifier: GPL-2.0 OR MIT
+ *
+ * bridge_tests.c — userspace unit + race tests of the rocp2p bridge,
+ * driven through the fake IB core / fake KFD the way real OFED + amdkfd
+ * would drive it.  Covers the behaviors the reference could only test on
+ * real hardware (reference test surface: /root/reference/tests/
+ * amdp2ptest.c) plus the races it could not test at all.
+ */
+#define __ROCNR_SHIM__ 1
+#include "rocnr_shim_all.h"
+#include "fake_kfd.h"
+#include "fake_ibcore.h"
+#include "rocp2p_sg.h"
+
+#include <assert.h>
+#include <time.h>
+#include <unistd.h>
+
+#define CHECK(cond) do { \
+	if (!(cond)) { \
+		fprintf(stderr, "FAIL %s:%d: %s\n", __FILE__, __LINE__, #cond); \
+		exit(1); \
+	} \
+} while (0)
+
+#define MiB (1ULL << 20)
+#define GiB (1ULL << 30)
+
+static void check_balances(struct device **devs, int ndev)
+{
+	int i;
+
+	CHECK(rocnr_shim_pid_balance() == 0);
+	CHECK(rocnr_shim_alloc_balance() == 0);
+	CHECK(rocnr_shim_sg_balance() == 0);
+	CHECK(rocnr_shim_module_refcount() == 0);
+	CHECK(fake_kfd_live_pins() == 0);
+	CHECK(fake_kfd_bad_put_calls() == 0);
+	for (i = 0; i < ndev; i++)
+		CHECK(atomic64_read(&devs[i]->live_maps) == 0);
+}
+
+/* ---- rocnr_coalesce unit tests (array-backed iterator) ---- */
+struct arr_iter {
+	struct rocnr_seg_iter it;
+	const struct rocnr_seg *segs;
+	size_t n, i;
+};
+
+static int arr_next(struct rocnr_seg_iter *it, struct rocnr_seg *seg)
+{
+	struct arr_iter *a = (struct arr_iter *)it;
+
+	if (a->i >= a->n)
+		return 0;
+	*seg = a->segs[a->i++];
+	return 1;
+}
+
+static struct arr_iter mk_iter(const struct rocnr_seg *segs, size_t n)
+{
+	struct arr_iter a = { { arr_next, 0 }, segs, n, 0 };
+	return a;
+}
+
+struct collect_ctx {
+	struct rocnr_seg out[64];
+	size_t n;
+};
+
+static int collect(void *ctx, const struct rocnr_seg *seg)
+{
+	struct collect_ctx *c = ctx;
+
+	if (c->n >= 64)
+		return 1;
+	c->out[c->n++] = *seg;
+	return 0;
+}
+
+static void test_coalesce_unit(void)
+{
+	struct collect_ctx c;
+	struct arr_iter it;
+
+	{	/* empty */
+		it = mk_iter(NULL, 0);
+		CHECK(rocnr_coalesce_count(&it.it, 0) == 0);
+	}
+	{	/* adjacent merge + gap */
+		const struct rocnr_seg s[] = {
+			{ 0x1000, 0x1000 }, { 0x2000, 0x1000 },
+			{ 0x4000, 0x1000 }, { 0x5000, 0x1000 },
+		};
+		it = mk_iter(s, 4);
+		c.n = 0;
+		CHECK(rocnr_coalesce(&it.it, 0, collect, &c) == 2);
+		CHECK(c.out[0].addr == 0x1000 && c.out[0].len == 0x2000);
+		CHECK(c.out[1].addr == 0x4000 && c.out[1].len == 0x2000);
+	}
+	{	/* zero-length entries skipped */
+		const struct rocnr_seg s[] = {
+			{ 0x1000, 0 }, { 0x1000, 0x1000 }, { 0x2000, 0 },
+			{ 0x2000, 0x1000 },
+		};
+		it = mk_iter(s, 4);
+		CHECK(rocnr_coalesce_count(&it.it, 0) == 1);
+	}
+	{	/* max_seg split: 10 pages merged then split at 3 pages */
+		struct rocnr_seg s[10];
+		int i;
+
+		for (i = 0; i < 10; i++) {
+			s[i].addr = 0x100000 + (rocnr_u64)i * 0x1000;
+			s[i].len = 0x1000;
+		}
+		it = mk_iter(s, 10);
+		c.n = 0;
+		CHECK(rocnr_coalesce(&it.it, 0x3000, collect, &c) == 4);
+		CHECK(c.out[0].len == 0x3000 && c.out[3].len == 0x1000);
+	}
+	{	/* emit abort propagates */
+		struct rocnr_seg s[80];
+		int i;
+
+		for (i = 0; i < 80; i++) {
+			s[i].addr = 0x2000000 + (rocnr_u64)i * 0x2000;
+			s[i].len = 0x1000;	/* all disjoint */
+		}
+		it = mk_iter(s, 80);
+		c.n = 0;
+		CHECK(rocnr_coalesce(&it.it, 0, collect, &c) == (size_t)-1);
+	}
+	printf("ok: coalesce unit\n");
+}
+
+/* ---- bridge flows ---- */
+static struct device dev1 = { .name = "hca0", .iova_offset = 0x10000000000ULL,
+			      .fail_after = -1 };
+static struct device dev2 = { .name = "hca1", .iova_offset = 0x20000000000ULL,
+			      .fail_after = -1 };
+static struct device *devs[] = { &dev1, &dev2 };
+
+static void test_not_gpu(void)
+{
+	struct fake_ib_mr *mr = NULL;
+
+	CHECK(fake_ib_reg_mr(0x1234000, 4096, &dev1, &mr) == -ENODEV);
+	check_balances(devs, 2);
+	printf("ok: not-gpu fallback\n");
+}
+
+static void test_lifecycle(void)
+{
+	uint64_t va = fake_kfd_alloc(64 * MiB, 0);
+	struct fake_ib_mr *mr = NULL;
+	uint64_t total = 0;
+	struct scatterlist *sg;
+	int i;
+
+	CHECK(fake_ib_reg_mr(va, 64 * MiB, &dev1, &mr) == 0);
+	CHECK(mr->page_size == 2 * MiB);
+	/* 32 bus-contiguous 2 MiB chunks must coalesce to ONE segment */
+	CHECK(mr->nmap == 1);
+	for_each_sg(mr->sgt.sgl, sg, mr->sgt.nents, i)
+		total += sg_dma_len(sg);
+	CHECK(total == 64 * MiB);
+	/* simulated IOMMU applied */
+	CHECK(sg_dma_address(mr->sgt.sgl) >= dev1.iova_offset);
+
+	CHECK(fake_ib_dereg_mr(mr) == 0);
+	free(mr);
+	fake_kfd_free(va);
+	check_balances(devs, 2);
+	printf("ok: lifecycle + full coalesce\n");
+}
+
+static void test_fragmented(void)
+{
+	/* hole after every 4th chunk: 32 chunks -> 8 runs */
+	uint64_t va = fake_kfd_alloc(64 * MiB, 4);
+	struct fake_ib_mr *mr = NULL;
+
+	CHECK(fake_ib_reg_mr(va, 64 * MiB, &dev1, &mr) == 0);
+	CHECK(mr->nmap == 8);
+	CHECK(fake_ib_dereg_mr(mr) == 0);
+	free(mr);
+	fake_kfd_free(va);
+	check_balances(devs, 2);
+	printf("ok: fragmented pin coalesces per run\n");
+}
+
+static void test_max_seg(void)
+{
+	uint64_t va = fake_kfd_alloc(64 * MiB, 0);
+	struct device small = { .name = "small", .max_seg = 8 * MiB,
+				.fail_after = -1 };
+	struct fake_ib_mr *mr = NULL;
+
+	CHECK(fake_ib_reg_mr(va, 64 * MiB, &small, &mr) == 0);
+	CHECK(mr->nmap == 8);	/* 64 MiB / 8 MiB max_seg */
+	CHECK(fake_ib_dereg_mr(mr) == 0);
+	free(mr);
+	fake_kfd_free(va);
+	CHECK(atomic64_read(&small.live_maps) == 0);
+	check_balances(devs, 2);
+	printf("ok: device max_seg honored\n");
+}
+
+static void test_range_mismatch(void)
+{
+	uint64_t va = fake_kfd_alloc(8 * MiB, 0);
+	const struct peer_memory_client *cl = fake_ib_client();
+	void *ctx = NULL;
+	char name[IB_PEER_MEMORY_NAME_MAX];
+	struct sg_table sgt = { 0 };
+
+	CHECK(cl->acquire(va, 8 * MiB, NULL, name, &ctx) == 1);
+	CHECK(cl->get_pages(va + 4096, 8 * MiB, 1, 0, &sgt, ctx, 0) ==
+	      -EINVAL);
+	CHECK(cl->get_pages(va, 4 * MiB, 1, 0, &sgt, ctx, 0) == -EINVAL);
+	cl->release(ctx);
+	fake_kfd_free(va);
+	check_balances(devs, 2);
+	printf("ok: get_pages range mismatch rejected\n");
+}
+
+static void test_map_failure_unwinds(void)
+{
+	uint64_t va = fake_kfd_alloc(64 * MiB, 4);	/* 8 runs */
+	struct device flaky = { .name = "flaky", .fail_after = 3,
+				.iova_offset = 0 };
+	struct fake_ib_mr *mr = NULL;
+
+	CHECK(fake_ib_reg_mr(va, 64 * MiB, &flaky, &mr) != 0);
+	CHECK(atomic64_read(&flaky.live_maps) == 0);
+	fake_kfd_free(va);
+	check_balances(devs, 2);
+	printf("ok: dma_map failure unwinds cleanly\n");
+}
+
+static void test_two_devices(void)
+{
+	uint64_t va = fake_kfd_alloc(32 * MiB, 0);
+	struct fake_ib_mr *mr = NULL;
+	struct sg_table sgt2 = { 0 };
+	int nmap2 = 0;
+
+	CHECK(fake_ib_reg_mr(va, 32 * MiB, &dev1, &mr) == 0);
+	CHECK(fake_ib_mr_map_also(mr, &dev2, &sgt2, &nmap2) == 0);
+	CHECK(nmap2 == 1);
+	CHECK(sg_dma_address(sgt2.sgl) >= dev2.iova_offset);
+	CHECK(sg_dma_address(sgt2.sgl) != sg_dma_address(mr->sgt.sgl));
+	CHECK(fake_ib_mr_unmap_also(mr, &dev2, &sgt2) == 0);
+	CHECK(fake_ib_dereg_mr(mr) == 0);
+	free(mr);
+	fake_kfd_free(va);
+	check_balances(devs, 2);
+	printf("ok: two concurrent device mappings\n");
+}
+
+static void test_release_cleans_leftovers(void)
+{
+	uint64_t va = fake_kfd_alloc(16 * MiB, 0);
+	struct fake_ib_mr *mr = NULL;
+	struct sg_table sgt2 = { 0 };
+	int nmap2 = 0;
+
+	CHECK(fake_ib_reg_mr(va, 16 * MiB, &dev1, &mr) == 0);
+	CHECK(fake_ib_mr_map_also(mr, &dev2, &sgt2, &nmap2) == 0);
+	/* dereg WITHOUT unmapping dev2: release must clean it up */
+	CHECK(fake_ib_dereg_mr(mr) == 0);
+	CHECK(atomic64_read(&dev2.live_maps) == 0);
+	free(mr);
+	fake_kfd_free(va);
+	check_balances(devs, 2);
+	printf("ok: release cleans leftover mappings\n");
+}
+
+static void test_invalidate(void)
+{
+	uint64_t va = fake_kfd_alloc(16 * MiB, 0);
+	struct fake_ib_mr *mr = NULL;
+	long inv0 = fake_ib_invalidate_count();
+
+	CHECK(fake_ib_reg_mr(va, 16 * MiB, &dev1, &mr) == 0);
+	fake_kfd_free(va);	/* GPU frees under a live MR */
+	CHECK(fake_ib_invalidate_count() == inv0 + 1);
+	CHECK(mr->invalidated == 1);
+	CHECK(fake_kfd_bad_put_calls() == 0);
+	/* app's dereg after invalidation must be a safe no-op */
+	CHECK(fake_ib_dereg_mr(mr) == 0);
+	free(mr);
+	check_balances(devs, 2);
+	printf("ok: async invalidation tears down MR exactly once\n");
+}
+
+struct race_arg {
+	uint64_t va;
+	unsigned int delay_ns;
+};
+
+static void *race_free_thread(void *argp)
+{
+	struct race_arg *a = argp;
+	struct timespec ts = { 0, a->delay_ns };
+
+	nanosleep(&ts, NULL);
+	fake_kfd_free(a->va);
+	return NULL;
+}
+
+static void test_invalidate_race(void)
+{
+	enum { ITERS = 3000 };
+	unsigned int seed = 12345;
+	int i;
+
+	for (i = 0; i < ITERS; i++) {
+		struct fake_ib_mr *mr = NULL;
+		struct race_arg a;
+		pthread_t th;
+		struct timespec ts = { 0, 0 };
+
+		a.va = fake_kfd_alloc(4 * MiB, 0);
+		CHECK(fake_ib_reg_mr(a.va, 4 * MiB, &dev1, &mr) == 0);
+		a.delay_ns = rand_r(&seed) % 20000;
+		pthread_create(&th, NULL, race_free_thread, &a);
+		ts.tv_nsec = rand_r(&seed) % 20000;
+		nanosleep(&ts, NULL);
+		fake_ib_dereg_mr(mr);
+		pthread_join(th, NULL);
+		fake_ib_dereg_mr(mr);	/* idempotent */
+		free(mr);
+		CHECK(fake_kfd_bad_put_calls() == 0);
+	}
+	check_balances(devs, 2);
+	printf("ok: %d dereg-vs-invalidate races, no double-put, no leak\n",
+	       ITERS);
+}
+
+static void test_huge_pin(void)
+{
+	/* 64 GiB pin (288 GB HBM sizing): 32768 chunks, hole every 1024 */
+	uint64_t va = fake_kfd_alloc(64 * GiB, 1024);
+	struct fake_ib_mr *mr = NULL;
+
+	CHECK(fake_ib_reg_mr(va, 64 * GiB, &dev1, &mr) == 0);
+	/* 32 runs of 2 GiB, each under the 4 GiB default max_seg */
+	CHECK(mr->nmap == 32);
+	CHECK(fake_ib_dereg_mr(mr) == 0);
+	free(mr);
+	fake_kfd_free(va);
+	check_balances(devs, 2);
+	printf("ok: 64 GiB pin -> 32 sg entries\n");
+}
+
+int main(void)
+{
+	test_coalesce_unit();
+
+	CHECK(rocnr_shim_module_init() == 0);
+	CHECK(fake_ib_client() != NULL);
+
+	test_not_gpu();
+	test_lifecycle();
+	test_fragmented();
+	test_max_seg();
+	test_range_mismatch();
+	test_map_failure_unwinds();
+	test_two_devices();
+	test_release_cleans_leftovers();
+	test_invalidate();
+	test_invalidate_race();
+	test_huge_pin();
+
+	rocnr_shim_module_exit();
+	CHECK(fake_ib_client() == NULL);
+
+	printf("ALL BRIDGE TESTS PASSED\n");
+	return 0;
+}
