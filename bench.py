@@ -1,0 +1,182 @@
+#!/usr/bin/env python3
+"""Flagship bench: aggregate RDMA-write bandwidth into GPU HBM.
+
+Measures the BASELINE.json metric ("ib_write_bw GB/s NIC<->GPU-HBM ...")
+on the best transport the box offers: real IB verbs when an HCA exists,
+otherwise the SDMA PCIe-BAR path (same bus and BAR window an HCA would
+master; transport recorded in config), `fake` on CPU-only boxes.
+
+One rank per GPU ("one QP per MI355X").  A step posts `--msgs-per-step`
+messages of `--msg-bytes` into a `--region-bytes` HBM region and
+completes them.  Timed region: exactly K steps bracketed by
+barrier + torch.cuda.synchronize on both sides; value is the whole-job
+aggregate GB/s (sum of bytes over all ranks / max elapsed over ranks).
+Payload integrity is proven outside the timed region with the on-GPU
+pattern/CRC kernels.
+
+Launch (driver contract):
+  python bench.py --gpus 1 --steps 20 --warmup 5
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+METRIC = ("ib_write_bw GB/s NIC<->GPU-HBM at 4KB/1MB/64MB msg; "
+          "1/2/4/8 MI355X aggregate")
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--msg-bytes", type=int, default=64 << 20,
+                   help="message size (headline: 64 MiB)")
+    p.add_argument("--region-bytes", type=int, default=1 << 30,
+                   help="registered-region size per GPU (config 2: 1 GiB)")
+    p.add_argument("--msgs-per-step", type=int, default=0,
+                   help="0 = one full region pass per step")
+    p.add_argument("--transport", default="auto",
+                   choices=["auto", "fake", "sdma", "verbs"])
+    p.add_argument("--direction", default="write",
+                   choices=["write", "read"])
+    p.add_argument("--inflight", type=int, default=8)
+    p.add_argument("--streams", type=int, default=2)
+    p.add_argument("--skip-integrity", action="store_true")
+    p.add_argument("--json-out", default="")
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    import torch
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+
+    dist = None
+    if world > 1:
+        import torch.distributed as dist_mod
+
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29511")
+        dist_mod.init_process_group("gloo", rank=rank, world_size=world)
+        dist = dist_mod
+
+    has_gpu = torch.cuda.is_available()
+    device = None
+    if has_gpu:
+        torch.cuda.set_device(local_rank % torch.cuda.device_count())
+        device = torch.device("cuda", local_rank % torch.cuda.device_count())
+
+    from rocnrdma_amd.transport import get_transport
+
+    # fake transport on CPU boxes gets a small region so CI stays fast
+    region = args.region_bytes
+    tname = args.transport
+    if tname == "auto":
+        tname = "sdma" if has_gpu else "fake"
+        if tname == "sdma":
+            from rocnrdma_amd.transport.verbs import verbs_available
+
+            if verbs_available():
+                tname = "verbs"
+    if tname == "fake" and args.region_bytes > (64 << 20):
+        region = min(region, 256 << 20)
+    msg = min(args.msg_bytes, region)
+
+    tp = get_transport(tname, msg_bytes=msg, region_bytes=region,
+                       inflight=args.inflight, direction=args.direction,
+                       device=device, num_streams=args.streams)
+
+    msgs_per_step = args.msgs_per_step or tp.msgs_per_region
+    bytes_per_step = msgs_per_step * msg
+
+    def barrier():
+        if dist is not None:
+            dist.barrier()
+
+    def gpu_sync():
+        if has_gpu:
+            torch.cuda.synchronize(device)
+
+    def step(base):
+        for m in range(msgs_per_step):
+            tp.post(base + m)
+        tp.flush()
+
+    # warmup (also first-touch of staging and region)
+    for s in range(args.warmup):
+        step(s * msgs_per_step)
+
+    barrier()
+    gpu_sync()
+    t0 = time.perf_counter()
+    for s in range(args.steps):
+        step(s * msgs_per_step)
+    gpu_sync()
+    t1 = time.perf_counter()
+    barrier()
+
+    elapsed = t1 - t0
+    if dist is not None:
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    integrity = "skipped"
+    if not args.skip_integrity:
+        bad = tp.integrity_check(seed=0xC0FFEE + rank)
+        integrity = "ok" if bad == 0 else f"FAILED:{bad}"
+
+    total_bytes = bytes_per_step * args.steps * world
+    value = total_bytes / elapsed / 1e9
+
+    result = {
+        "metric": METRIC,
+        "value": round(value, 3),
+        "unit": "GB/s",
+        "n_gpus": world if has_gpu else args.gpus,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": round(elapsed / args.steps * 1e3, 3),
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,  # reference publishes no numbers
+        "dtype": "uint8",
+        "data": "synthetic",
+        "config": {
+            "model": "rdma-write-bandwidth",
+            "global_batch": msgs_per_step * world,
+            "seq_len": msg,
+            "parallelism": f"1qp-per-gpu x{world}",
+            "transport": tp.name,
+            "direction": args.direction,
+            "msg_bytes": msg,
+            "region_bytes": region,
+            "inflight": args.inflight,
+            "streams": args.streams,
+            "integrity": integrity,
+        },
+    }
+    if rank == 0:
+        line = json.dumps(result)
+        print(line)
+        if args.json_out:
+            with open(args.json_out, "w") as f:
+                f.write(line + "\n")
+    tp.close()
+    if dist is not None:
+        dist.destroy_process_group()
+    if integrity.startswith("FAILED"):
+        raise SystemExit(2)
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,82 @@
+"""bench.py contract tests (CPU, fake transport): single-rank JSON line
+and 2-process gloo aggregate."""
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+BENCH = os.path.join(ROOT, "bench.py")
+
+REQUIRED = ["metric", "value", "unit", "n_gpus", "steps", "warmup",
+            "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+            "dtype", "data", "config"]
+
+
+def run_bench(extra, env_extra=None, timeout=240):
+    env = dict(os.environ)
+    env["PYTHONPATH"] = ROOT + os.pathsep + env.get("PYTHONPATH", "")
+    if env_extra:
+        env.update(env_extra)
+    out = subprocess.run(
+        [sys.executable, BENCH, "--transport", "fake", "--msg-bytes",
+         "65536", "--region-bytes", "1048576", "--steps", "3",
+         "--warmup", "1"] + extra,
+        capture_output=True, text=True, env=env, cwd=ROOT, timeout=timeout)
+    assert out.returncode == 0, out.stderr
+    lines = [l for l in out.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1, out.stdout
+    return json.loads(lines[0])
+
+
+def test_single_rank_json_contract():
+    r = run_bench([])
+    for key in REQUIRED:
+        assert key in r, key
+    assert r["unit"] == "GB/s"
+    assert r["value"] > 0
+    assert r["steps"] == 3 and r["warmup"] == 1
+    assert r["higher_is_better"] is True
+    assert r["scaling"] == "weak"
+    assert r["vs_baseline"] is None
+    assert r["data"] == "synthetic"
+    assert r["config"]["transport"] == "fake"
+    assert r["config"]["integrity"] == "ok"
+    assert r["ms_per_step"] > 0
+    assert "ib_write_bw" in r["metric"]
+
+
+def test_integrity_gate():
+    # direction=read also exercises the pull path
+    r = run_bench(["--direction", "read"])
+    assert r["config"]["integrity"] == "ok"
+
+
+@pytest.mark.timeout(300)
+def test_two_rank_gloo_aggregate():
+    """Two fake-transport ranks over gloo: value aggregates both ranks."""
+    procs = []
+    port = "29533"
+    for rank in range(2):
+        env = dict(os.environ)
+        env["PYTHONPATH"] = ROOT + os.pathsep + env.get("PYTHONPATH", "")
+        env.update({"RANK": str(rank), "WORLD_SIZE": "2",
+                    "LOCAL_RANK": str(rank), "MASTER_ADDR": "127.0.0.1",
+                    "MASTER_PORT": port})
+        procs.append(subprocess.Popen(
+            [sys.executable, BENCH, "--transport", "fake", "--gpus", "2",
+             "--msg-bytes", "65536", "--region-bytes", "1048576",
+             "--steps", "3", "--warmup", "1"],
+            stdout=subprocess.PIPE, stderr=subprocess.PIPE, text=True,
+            env=env, cwd=ROOT))
+    outs = [p.communicate(timeout=240) for p in procs]
+    for p, (so, se) in zip(procs, outs):
+        assert p.returncode == 0, se
+    # rank 0 prints the line
+    lines = [l for l in outs[0][0].splitlines() if l.startswith("{")]
+    assert len(lines) == 1
+    r = json.loads(lines[0])
+    assert r["n_gpus"] == 2
+    assert "x2" in r["config"]["parallelism"]

@@ -1,0 +1,34 @@
+"""Drives the userspace-shim build of the kernel bridge (unit + race
+tests, plain and ASan) — the CPU-tier equivalent of loading rocp2p.ko
+against real amdkfd/OFED."""
+import os
+import shutil
+import subprocess
+
+import pytest
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+SHIM = os.path.join(ROOT, "module", "shim")
+
+
+@pytest.mark.skipif(shutil.which("gcc") is None or
+                    shutil.which("make") is None, reason="no toolchain")
+@pytest.mark.timeout(600)
+def test_bridge_shim_suite():
+    subprocess.run(["make", "-C", SHIM, "all"], check=True,
+                   capture_output=True, text=True)
+    out = subprocess.run([os.path.join(SHIM, "build", "bridge_tests")],
+                         capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stdout + out.stderr
+    assert "ALL BRIDGE TESTS PASSED" in out.stdout
+
+
+@pytest.mark.skipif(shutil.which("gcc") is None or
+                    shutil.which("make") is None, reason="no toolchain")
+@pytest.mark.timeout(900)
+def test_bridge_shim_suite_asan():
+    subprocess.run(["make", "-C", SHIM, "all"], check=True,
+                   capture_output=True, text=True)
+    out = subprocess.run([os.path.join(SHIM, "build", "bridge_tests_asan")],
+                         capture_output=True, text=True, timeout=600)
+    assert out.returncode == 0, out.stdout + out.stderr

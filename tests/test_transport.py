@@ -1,0 +1,56 @@
+"""CPU tests of the transport layer via the fake (loopback) backend —
+BASELINE config 1."""
+import numpy as np
+import pytest
+
+from rocnrdma_amd.transport import available_transports, get_transport
+from rocnrdma_amd.utils import pattern
+
+
+def test_available_contains_fake():
+    assert "fake" in available_transports()
+
+
+def test_bad_geometry_rejected():
+    with pytest.raises(ValueError):
+        get_transport("fake", msg_bytes=4096, region_bytes=10000)
+    with pytest.raises(ValueError):
+        get_transport("fake", msg_bytes=4096, region_bytes=8192,
+                      direction="sideways")
+    with pytest.raises(ValueError):
+        get_transport("nope", msg_bytes=4096, region_bytes=8192)
+
+
+@pytest.mark.parametrize("direction", ["write", "read"])
+def test_fake_integrity(direction):
+    tp = get_transport("fake", msg_bytes=4096, region_bytes=64 * 1024,
+                       direction=direction)
+    assert tp.msgs_per_region == 16
+    assert tp.integrity_check(seed=42) == 0
+
+
+def test_fake_write_moves_bytes():
+    tp = get_transport("fake", msg_bytes=1024, region_bytes=4096)
+    tp.staging[0][:] = 7
+    tp.post(0)
+    tp.flush()
+    assert (tp.region[:1024] == 7).all()
+    assert (tp.region[1024:] == 0).all()
+
+
+def test_fake_detects_corruption():
+    tp = get_transport("fake", msg_bytes=4096, region_bytes=16 * 4096)
+    assert tp.integrity_check(seed=9) == 0
+    # corrupt one destination word post-hoc: a rerun must transfer fresh
+    # payloads and still pass; direct verification of a corrupted region
+    # is the GPU kernels' job (test_gpu_kernels.py)
+    tp.region[100] ^= 0xFF
+    ref = pattern.fill_reference(tp.region_bytes, 9)
+    assert int(np.count_nonzero(
+        tp.region.view(np.uint64) != ref.view(np.uint64))) == 1
+
+
+def test_inflight_clamped():
+    tp = get_transport("fake", msg_bytes=4096, region_bytes=8192,
+                       inflight=64)
+    assert tp.inflight == 2
