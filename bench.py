@@ -45,7 +45,9 @@ def parse_args():
                    choices=["auto", "fake", "sdma", "verbs"])
     p.add_argument("--direction", default="write",
                    choices=["write", "read"])
-    p.add_argument("--inflight", type=int, default=8)
+    p.add_argument("--inflight", type=int, default=0,
+                   help="0 = transport-chosen (kernel engine sizes its "
+                        "WQE ring for ~64 MiB of pinned staging)")
     p.add_argument("--streams", type=int, default=2)
     p.add_argument("--skip-integrity", action="store_true")
     p.add_argument("--json-out", default="")

@@ -56,7 +56,7 @@ def run_point(tp, target_secs: float = 1.0, min_msgs: int = 8) -> dict:
 def run_sweep(transport: str = "auto", region_bytes: int = 1 << 30,
               sizes=None, directions=("write", "read"),
               target_secs: float = 1.0, device=None,
-              num_streams: int = 2, inflight: int = 8) -> list[dict]:
+              num_streams: int = 2, inflight: int | None = 0) -> list[dict]:
     from rocnrdma_amd.transport import get_transport
 
     rows = []
@@ -82,7 +82,7 @@ def main():
     ap.add_argument("--region-bytes", type=int, default=1 << 30)
     ap.add_argument("--secs", type=float, default=1.0)
     ap.add_argument("--streams", type=int, default=2)
-    ap.add_argument("--inflight", type=int, default=8)
+    ap.add_argument("--inflight", type=int, default=0)
     ap.add_argument("--sizes", default="")
     ap.add_argument("--out", default="")
     args = ap.parse_args()

@@ -46,3 +46,16 @@ def crc32_pages(buf: torch.Tensor) -> torch.Tensor:
 def copy_(dst: torch.Tensor, src: torch.Tensor) -> None:
     """Streaming 16 B/lane device copy (bandwidth ceiling probe)."""
     _require().copy_(dst, src)
+
+
+def gather_(region: torch.Tensor, dst_offs: torch.Tensor,
+            src_addrs: torch.Tensor, msg_bytes: int) -> None:
+    """Batched message engine: host-pinned sources -> HBM region offsets
+    (one launch retires the whole batch — the NIC-WQE analog)."""
+    _require().gather_(region, dst_offs, src_addrs, msg_bytes)
+
+
+def scatter_(region: torch.Tensor, src_offs: torch.Tensor,
+             dst_addrs: torch.Tensor, msg_bytes: int) -> None:
+    """Batched message engine: HBM region offsets -> host-pinned dests."""
+    _require().scatter_(region, src_offs, dst_addrs, msg_bytes)

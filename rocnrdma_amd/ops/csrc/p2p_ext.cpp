@@ -58,6 +58,40 @@ torch::Tensor crc32_pages(torch::Tensor buf) {
   return out;
 }
 
+void check_u64_dev(const torch::Tensor& t, const char* who) {
+  TORCH_CHECK(t.is_cuda() && t.is_contiguous() &&
+                  t.scalar_type() == torch::kInt64,
+              who, ": descriptor tensor must be contiguous int64 on GPU");
+}
+
+void gather_(torch::Tensor region, torch::Tensor dst_offs,
+             torch::Tensor src_addrs, int64_t msg_bytes) {
+  check_buf(region, "gather_");
+  check_u64_dev(dst_offs, "gather_");
+  check_u64_dev(src_addrs, "gather_");
+  TORCH_CHECK(dst_offs.numel() == src_addrs.numel(), "gather_: n mismatch");
+  auto stream = at::cuda::getCurrentCUDAStream();
+  HIP_OK(rocp2p_gather(region.data_ptr(),
+                       (const uint64_t*)dst_offs.data_ptr<int64_t>(),
+                       (const uint64_t*)src_addrs.data_ptr<int64_t>(),
+                       (uint64_t)msg_bytes, (uint32_t)dst_offs.numel(),
+                       stream.stream()));
+}
+
+void scatter_(torch::Tensor region, torch::Tensor src_offs,
+              torch::Tensor dst_addrs, int64_t msg_bytes) {
+  check_buf(region, "scatter_");
+  check_u64_dev(src_offs, "scatter_");
+  check_u64_dev(dst_addrs, "scatter_");
+  TORCH_CHECK(src_offs.numel() == dst_addrs.numel(), "scatter_: n mismatch");
+  auto stream = at::cuda::getCurrentCUDAStream();
+  HIP_OK(rocp2p_scatter(region.data_ptr(),
+                        (const uint64_t*)src_offs.data_ptr<int64_t>(),
+                        (const uint64_t*)dst_addrs.data_ptr<int64_t>(),
+                        (uint64_t)msg_bytes, (uint32_t)src_offs.numel(),
+                        stream.stream()));
+}
+
 void copy_(torch::Tensor dst, torch::Tensor src) {
   check_buf(dst, "copy_");
   check_buf(src, "copy_");
@@ -76,4 +110,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("verify", &verify, "count words deviating from the pattern");
   m.def("crc32_pages", &crc32_pages, "zlib CRC32 of each 4 KiB page");
   m.def("copy_", &copy_, "streaming device copy dst <- src");
+  m.def("gather_", &gather_,
+        "batched message engine: host-pinned srcs -> HBM region offsets");
+  m.def("scatter_", &scatter_,
+        "batched message engine: HBM region offsets -> host-pinned dsts");
 }

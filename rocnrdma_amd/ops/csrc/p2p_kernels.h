@@ -31,4 +31,16 @@ hipError_t rocp2p_crc32_pages(const void* buf, uint64_t npages,
 hipError_t rocp2p_copy(void* dst, const void* src, uint64_t nbytes,
                        hipStream_t stream);
 
+// GPU-driven batched message engine (the NIC-WQE analog: one launch
+// retires a whole queue of posted messages).  src_addrs/dst_offs are
+// device-visible u64 arrays of n entries; every message is msg_bytes
+// (16-byte multiple).  gather: host-pinned sources -> HBM region.
+// scatter: HBM region -> host-pinned destinations.
+hipError_t rocp2p_gather(void* dst_base, const uint64_t* d_dst_offs,
+                         const uint64_t* d_src_addrs, uint64_t msg_bytes,
+                         uint32_t n, hipStream_t stream);
+hipError_t rocp2p_scatter(const void* src_base, const uint64_t* d_src_offs,
+                          const uint64_t* d_dst_addrs, uint64_t msg_bytes,
+                          uint32_t n, hipStream_t stream);
+
 }  // extern "C"

@@ -94,6 +94,49 @@ __global__ void k_copy(uint4* __restrict__ dst, const uint4* __restrict__ src,
 }
 
 // ---------------------------------------------------------------------
+// Batched message engine.  A real HCA retires posted WQEs with its DMA
+// hardware regardless of message size; the host-API (hipMemcpyAsync)
+// path costs ~10 us per message, capping 4 KB messages near 0.3 GB/s.
+// Here one kernel launch retires the whole posted batch: lanes read
+// host-pinned staging straight over PCIe (fine-grained zero-copy) and
+// store to the HBM region, 16 B/lane, many messages per launch — so the
+// small-message lines of the ib_write_bw sweep are PCIe-bound, not
+// launch-bound.  msg_bytes is uniform per batch (transport invariant).
+
+__global__ void k_gather(uint8_t* __restrict__ dst_base,
+                         const uint64_t* __restrict__ dst_offs,
+                         const uint64_t* __restrict__ src_addrs,
+                         uint64_t vecs_per_msg, uint64_t total_vecs,
+                         uint32_t n_msgs) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t v = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x;
+       v < total_vecs; v += stride) {
+    uint64_t msg = v / vecs_per_msg;
+    uint64_t idx = v - msg * vecs_per_msg;
+    const uint4* src = reinterpret_cast<const uint4*>(src_addrs[msg]);
+    uint4* dst = reinterpret_cast<uint4*>(dst_base + dst_offs[msg]);
+    dst[idx] = src[idx];
+  }
+}
+
+__global__ void k_scatter(const uint8_t* __restrict__ src_base,
+                          const uint64_t* __restrict__ src_offs,
+                          const uint64_t* __restrict__ dst_addrs,
+                          uint64_t vecs_per_msg, uint64_t total_vecs,
+                          uint32_t n_msgs) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t v = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x;
+       v < total_vecs; v += stride) {
+    uint64_t msg = v / vecs_per_msg;
+    uint64_t idx = v - msg * vecs_per_msg;
+    const uint4* src =
+        reinterpret_cast<const uint4*>(src_base + src_offs[msg]);
+    uint4* dst = reinterpret_cast<uint4*>(dst_addrs[msg]);
+    dst[idx] = src[idx];
+  }
+}
+
+// ---------------------------------------------------------------------
 // CRC32 (zlib polynomial 0xEDB88320, init/xorout 0xFFFFFFFF)
 
 __constant__ uint32_t c_crc_tab[8][256];   // slice-by-8
@@ -247,6 +290,38 @@ extern "C" hipError_t rocp2p_copy(void* dst, const void* src, uint64_t nbytes,
   uint32_t grid = stream_grid(nvec, 256);
   hipLaunchKernelGGL(k_copy, dim3(grid), dim3(256), 0, stream, (uint4*)dst,
                      (const uint4*)src, nvec);
+  return hipGetLastError();
+}
+
+extern "C" hipError_t rocp2p_gather(void* dst_base,
+                                    const uint64_t* d_dst_offs,
+                                    const uint64_t* d_src_addrs,
+                                    uint64_t msg_bytes, uint32_t n,
+                                    hipStream_t stream) {
+  if (msg_bytes % 16 || !msg_bytes) return hipErrorInvalidValue;
+  if (!n) return hipSuccess;
+  uint64_t vecs_per_msg = msg_bytes / 16;
+  uint64_t total = vecs_per_msg * n;
+  uint32_t grid = stream_grid(total, 256);
+  hipLaunchKernelGGL(k_gather, dim3(grid), dim3(256), 0, stream,
+                     (uint8_t*)dst_base, d_dst_offs, d_src_addrs,
+                     vecs_per_msg, total, n);
+  return hipGetLastError();
+}
+
+extern "C" hipError_t rocp2p_scatter(const void* src_base,
+                                     const uint64_t* d_src_offs,
+                                     const uint64_t* d_dst_addrs,
+                                     uint64_t msg_bytes, uint32_t n,
+                                     hipStream_t stream) {
+  if (msg_bytes % 16 || !msg_bytes) return hipErrorInvalidValue;
+  if (!n) return hipSuccess;
+  uint64_t vecs_per_msg = msg_bytes / 16;
+  uint64_t total = vecs_per_msg * n;
+  uint32_t grid = stream_grid(total, 256);
+  hipLaunchKernelGGL(k_scatter, dim3(grid), dim3(256), 0, stream,
+                     (const uint8_t*)src_base, d_src_offs, d_dst_addrs,
+                     vecs_per_msg, total, n);
   return hipGetLastError();
 }
 

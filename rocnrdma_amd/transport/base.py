@@ -20,12 +20,14 @@ import abc
 class Transport(abc.ABC):
     name = "base"
 
-    def __init__(self, msg_bytes: int, region_bytes: int, inflight: int = 8,
-                 direction: str = "write", **_):
+    def __init__(self, msg_bytes: int, region_bytes: int,
+                 inflight: int | None = 8, direction: str = "write", **_):
         if region_bytes % msg_bytes:
             raise ValueError("region must be a multiple of msg size")
         if direction not in ("write", "read"):
             raise ValueError(direction)
+        if not inflight:
+            inflight = 8
         self.msg_bytes = msg_bytes
         self.region_bytes = region_bytes
         self.inflight = max(1, min(inflight, region_bytes // msg_bytes))

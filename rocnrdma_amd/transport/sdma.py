@@ -1,15 +1,24 @@
-"""SDMA transport: the PCIe BAR data path on a GPU-only box.
+"""SDMA/PCIe transport: the BAR data path on a GPU-only box.
 
 An HCA doing RDMA WRITE into GPU HBM is a PCIe bus master writing into
-the GPU's BAR aperture; on a box without an HCA the closest measurable
-equivalent of that path is the GPU's own SDMA engines moving data
-between host-pinned memory and HBM across the same PCIe Gen5 x16 link
-(~63 GB/s spec ceiling).  One transport instance = one "QP"
-analog: a set of HIP streams (SDMA channels) + pinned staging + a large
-HBM destination region, on one GPU.
+the GPU's BAR aperture; on a box without an HCA the measurable
+equivalent of that path is moving data between host-pinned memory and
+HBM across the same PCIe Gen5 x16 link (~63 GB/s spec).
+One transport instance = one "QP" analog on one GPU.
 
-Integrity is proven on-GPU: pattern payloads staged from the host are
-verified in HBM by the verify/CRC kernels (zero host readback).
+Two engines, mirroring how a NIC actually retires work:
+
+- "stream": hipMemcpyAsync per message on round-robin HIP streams (SDMA
+  engines).  ~10 us host cost per message — fine >= 4 MiB, hopeless at
+  4 KiB (measured 0.3 GB/s).
+- "kernel": doorbell semantics.  post() appends a WQE (two u64 writes
+  into a pinned descriptor ring); flush() rings the doorbell — one
+  gather/scatter kernel launch retires the whole batch, lanes reading
+  host-pinned staging directly over PCIe (fine-grained zero-copy).
+  Small-message bandwidth becomes PCIe-bound instead of launch-bound.
+
+"auto" picks kernel below 4 MiB messages.  Integrity is proven on-GPU
+(verify/CRC kernels) — zero host readback for write direction.
 """
 from __future__ import annotations
 
@@ -18,41 +27,97 @@ import torch
 from ..utils import pattern
 from .base import Transport
 
+_KERNEL_THRESHOLD = 4 << 20
+_STAGING_TARGET = 64 << 20  # pinned staging budget for small messages
+
 
 class SdmaTransport(Transport):
     name = "sdma"
 
     def __init__(self, msg_bytes: int, region_bytes: int, device=None,
-                 num_streams: int = 2, **kw):
-        super().__init__(msg_bytes, region_bytes, **kw)
+                 num_streams: int = 2, engine: str = "auto",
+                 inflight: int | None = None, **kw):
         if not torch.cuda.is_available():
             raise RuntimeError("sdma transport requires a GPU")
+        if engine == "auto":
+            engine = "kernel" if msg_bytes < _KERNEL_THRESHOLD else "stream"
+        self.engine = engine
+        if inflight in (None, 0):
+            if engine == "kernel":
+                inflight = max(8, min(region_bytes // msg_bytes,
+                                      _STAGING_TARGET // msg_bytes, 16384))
+            else:
+                inflight = 8
+        super().__init__(msg_bytes, region_bytes, inflight=inflight, **kw)
         self.device = torch.device(device or "cuda")
         with torch.cuda.device(self.device):
             self.streams = [torch.cuda.Stream(self.device)
-                            for _ in range(num_streams)]
+                            for _ in range(max(1, num_streams))]
+        # one contiguous pinned staging area, sliced into slots
+        self.staging_flat = torch.empty(self.inflight * msg_bytes,
+                                        dtype=torch.uint8, pin_memory=True)
         self.staging = [
-            torch.empty(msg_bytes, dtype=torch.uint8, pin_memory=True)
-            for _ in range(self.inflight)
+            self.staging_flat[i * msg_bytes:(i + 1) * msg_bytes]
+            for i in range(self.inflight)
         ]
         self.region = torch.zeros(region_bytes, dtype=torch.uint8,
                                   device=self.device)
+        if self.engine == "kernel":
+            base = self.staging_flat.data_ptr()
+            self._slot_addr = [base + i * msg_bytes
+                               for i in range(self.inflight)]
+            # WQE ring: [0] = region offset, [1] = staging address
+            self._desc_pin = torch.empty((2, self.inflight),
+                                         dtype=torch.int64, pin_memory=True)
+            self._desc_np = self._desc_pin.numpy()
+            self._desc_dev = torch.empty((2, self.inflight),
+                                         dtype=torch.int64,
+                                         device=self.device)
+            self._pending = 0
 
+    # -- data plane ----------------------------------------------------
     def post(self, i: int) -> None:
-        slot = self.staging[i % self.inflight]
         off = (i % self.msgs_per_region) * self.msg_bytes
-        dst = self.region[off : off + self.msg_bytes]
+        slot = i % self.inflight
+        if self.engine == "kernel":
+            if self._pending >= self.inflight:
+                self.flush()
+            k = self._pending
+            self._desc_np[0, k] = off
+            self._desc_np[1, k] = self._slot_addr[slot]
+            self._pending = k + 1
+            return
         stream = self.streams[i % len(self.streams)]
+        dst = self.region[off:off + self.msg_bytes]
         with torch.cuda.stream(stream):
             if self.direction == "write":
-                dst.copy_(slot, non_blocking=True)
+                dst.copy_(self.staging[slot], non_blocking=True)
             else:
-                slot.copy_(dst, non_blocking=True)
+                self.staging[slot].copy_(dst, non_blocking=True)
 
     def flush(self) -> None:
+        if self.engine == "kernel":
+            n = self._pending
+            if n:
+                from .. import ops
+
+                self._pending = 0
+                stream = self.streams[0]
+                with torch.cuda.stream(stream):
+                    self._desc_dev[:, :n].copy_(self._desc_pin[:, :n],
+                                                non_blocking=True)
+                    if self.direction == "write":
+                        ops.gather_(self.region, self._desc_dev[0, :n],
+                                    self._desc_dev[1, :n], self.msg_bytes)
+                    else:
+                        ops.scatter_(self.region, self._desc_dev[0, :n],
+                                     self._desc_dev[1, :n], self.msg_bytes)
+            self.streams[0].synchronize()
+            return
         for s in self.streams:
             s.synchronize()
 
+    # -- integrity plane ----------------------------------------------
     def integrity_check(self, seed: int) -> int:
         from .. import ops
 
@@ -65,16 +130,14 @@ class SdmaTransport(Transport):
                 for j in range(i, i + burst):
                     off = j * self.msg_bytes
                     self.staging[j % self.inflight].copy_(
-                        ref_t[off : off + self.msg_bytes])
+                        ref_t[off:off + self.msg_bytes])
                     self.post(j)
                 self.flush()  # slots reused next burst: must complete
                 i += burst
             # on-GPU verification — no host readback
             return int(ops.verify(self.region, seed))
         # read: pattern HBM with the fill kernel, pull to host, check there
-        from .. import ops as _ops
-
-        _ops.fill_(self.region, seed)
+        ops.fill_(self.region, seed)
         torch.cuda.synchronize(self.device)
         bad = 0
         i = 0
@@ -86,7 +149,7 @@ class SdmaTransport(Transport):
             for j in range(i, i + burst):
                 off = j * self.msg_bytes
                 got = self.staging[j % self.inflight].numpy()
-                want = ref[off : off + self.msg_bytes]
+                want = ref[off:off + self.msg_bytes]
                 bad += int((got.view("u8") != want.view("u8")).sum())
             i += burst
         return bad

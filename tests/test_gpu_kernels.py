@@ -81,23 +81,50 @@ def test_copy_kernel(dev):
     assert torch.equal(dst, src)
 
 
+@pytest.mark.parametrize("engine", ["stream", "kernel"])
 @pytest.mark.parametrize("direction", ["write", "read"])
-def test_sdma_transport_integrity(dev, direction):
+def test_sdma_transport_integrity(dev, direction, engine):
     from rocnrdma_amd.transport import get_transport
 
     tp = get_transport("sdma", msg_bytes=1 << 20, region_bytes=32 << 20,
-                       device=dev, direction=direction)
+                       device=dev, direction=direction, engine=engine)
     assert tp.integrity_check(seed=1234) == 0
     tp.close()
 
 
-def test_sdma_transport_4kb_messages(dev):
+@pytest.mark.parametrize("direction", ["write", "read"])
+def test_sdma_transport_4kb_messages(dev, direction):
     from rocnrdma_amd.transport import get_transport
 
     tp = get_transport("sdma", msg_bytes=4096, region_bytes=1 << 20,
-                       device=dev)
+                       device=dev, direction=direction)
+    assert tp.engine == "kernel"
+    assert tp.inflight == 256  # full region fits the WQE ring
     assert tp.integrity_check(seed=77) == 0
     tp.close()
+
+
+def test_gather_kernel_matches_reference(dev):
+    """Direct gather_ use: scattered host-pinned slots -> chosen HBM
+    offsets, compared against a CPU-composed reference."""
+    import rocnrdma_amd.ops as ops
+
+    msg = 8192
+    n = 37
+    staging = torch.randint(0, 256, (n * msg,), dtype=torch.uint8,
+                            pin_memory=True)
+    region = torch.zeros(n * msg, dtype=torch.uint8, device=dev)
+    perm = torch.randperm(n)
+    dst_offs = (perm * msg).to(dtype=torch.int64, device=dev)
+    src_addrs = torch.tensor(
+        [staging.data_ptr() + i * msg for i in range(n)],
+        dtype=torch.int64, device=dev)
+    ops.gather_(region, dst_offs, src_addrs, msg)
+    torch.cuda.synchronize()
+    got = region.cpu()
+    for i in range(n):
+        off = int(perm[i]) * msg
+        assert torch.equal(got[off:off + msg], staging[i * msg:(i + 1) * msg])
 
 
 def test_smoke_entry():
