@@ -109,8 +109,7 @@ def main():
             torch.cuda.synchronize(device)
 
     def step(base):
-        for m in range(msgs_per_step):
-            tp.post(base + m)
+        tp.post_many(base, msgs_per_step)
         tp.flush()
 
     # warmup (also first-touch of staging and region)

@@ -25,9 +25,7 @@ def run_point(tp, target_secs: float = 1.0, min_msgs: int = 8) -> dict:
             torch.cuda.synchronize()
 
     # warmup: one region pass (capped)
-    n_warm = min(tp.msgs_per_region, 64)
-    for i in range(n_warm):
-        tp.post(i)
+    tp.post_many(0, min(tp.msgs_per_region, max(64, tp.inflight)))
     sync()
 
     # calibrate burst so each timed burst is >= ~50 ms
@@ -36,9 +34,8 @@ def run_point(tp, target_secs: float = 1.0, min_msgs: int = 8) -> dict:
     t_end = time.perf_counter() + target_secs
     t0 = time.perf_counter()
     while time.perf_counter() < t_end:
-        for _ in range(burst):
-            tp.post(posted)
-            posted += 1
+        tp.post_many(posted, burst)
+        posted += burst
         tp.flush()
     sync()
     elapsed = time.perf_counter() - t0

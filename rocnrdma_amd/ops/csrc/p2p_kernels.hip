@@ -103,15 +103,18 @@ __global__ void k_copy(uint4* __restrict__ dst, const uint4* __restrict__ src,
 // small-message lines of the ib_write_bw sweep are PCIe-bound, not
 // launch-bound.  msg_bytes is uniform per batch (transport invariant).
 
+// vshift: log2(vecs_per_msg) when it is a power of two (uniform branch
+// avoids a 64-bit divide per 16-byte vector), else 0xffffffff.
 __global__ void k_gather(uint8_t* __restrict__ dst_base,
                          const uint64_t* __restrict__ dst_offs,
                          const uint64_t* __restrict__ src_addrs,
                          uint64_t vecs_per_msg, uint64_t total_vecs,
-                         uint32_t n_msgs) {
+                         uint32_t vshift) {
   uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
   for (uint64_t v = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x;
        v < total_vecs; v += stride) {
-    uint64_t msg = v / vecs_per_msg;
+    uint64_t msg = (vshift != 0xffffffffu) ? (v >> vshift)
+                                           : (v / vecs_per_msg);
     uint64_t idx = v - msg * vecs_per_msg;
     const uint4* src = reinterpret_cast<const uint4*>(src_addrs[msg]);
     uint4* dst = reinterpret_cast<uint4*>(dst_base + dst_offs[msg]);
@@ -123,11 +126,12 @@ __global__ void k_scatter(const uint8_t* __restrict__ src_base,
                           const uint64_t* __restrict__ src_offs,
                           const uint64_t* __restrict__ dst_addrs,
                           uint64_t vecs_per_msg, uint64_t total_vecs,
-                          uint32_t n_msgs) {
+                          uint32_t vshift) {
   uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
   for (uint64_t v = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x;
        v < total_vecs; v += stride) {
-    uint64_t msg = v / vecs_per_msg;
+    uint64_t msg = (vshift != 0xffffffffu) ? (v >> vshift)
+                                           : (v / vecs_per_msg);
     uint64_t idx = v - msg * vecs_per_msg;
     const uint4* src =
         reinterpret_cast<const uint4*>(src_base + src_offs[msg]);
@@ -303,9 +307,12 @@ extern "C" hipError_t rocp2p_gather(void* dst_base,
   uint64_t vecs_per_msg = msg_bytes / 16;
   uint64_t total = vecs_per_msg * n;
   uint32_t grid = stream_grid(total, 256);
+  uint32_t vshift = (vecs_per_msg & (vecs_per_msg - 1))
+                        ? 0xffffffffu
+                        : (uint32_t)__builtin_ctzll(vecs_per_msg);
   hipLaunchKernelGGL(k_gather, dim3(grid), dim3(256), 0, stream,
                      (uint8_t*)dst_base, d_dst_offs, d_src_addrs,
-                     vecs_per_msg, total, n);
+                     vecs_per_msg, total, vshift);
   return hipGetLastError();
 }
 
@@ -319,9 +326,12 @@ extern "C" hipError_t rocp2p_scatter(const void* src_base,
   uint64_t vecs_per_msg = msg_bytes / 16;
   uint64_t total = vecs_per_msg * n;
   uint32_t grid = stream_grid(total, 256);
+  uint32_t vshift = (vecs_per_msg & (vecs_per_msg - 1))
+                        ? 0xffffffffu
+                        : (uint32_t)__builtin_ctzll(vecs_per_msg);
   hipLaunchKernelGGL(k_scatter, dim3(grid), dim3(256), 0, stream,
                      (const uint8_t*)src_base, d_src_offs, d_dst_addrs,
-                     vecs_per_msg, total, n);
+                     vecs_per_msg, total, vshift);
   return hipGetLastError();
 }
 

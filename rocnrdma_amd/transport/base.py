@@ -42,6 +42,12 @@ class Transport(abc.ABC):
     def flush(self) -> None:
         """Complete every outstanding message."""
 
+    def post_many(self, start: int, n: int) -> None:
+        """Enqueue messages start..start+n-1 (backends may vectorize the
+        WQE writes; semantics identical to n post() calls)."""
+        for i in range(start, start + n):
+            self.post(i)
+
     # -- integrity plane (never timed) ---------------------------------
     @abc.abstractmethod
     def integrity_check(self, seed: int) -> int:

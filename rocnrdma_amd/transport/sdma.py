@@ -63,9 +63,12 @@ class SdmaTransport(Transport):
         self.region = torch.zeros(region_bytes, dtype=torch.uint8,
                                   device=self.device)
         if self.engine == "kernel":
+            import numpy as np
+
             base = self.staging_flat.data_ptr()
             self._slot_addr = [base + i * msg_bytes
                                for i in range(self.inflight)]
+            self._slot_addr_np = np.array(self._slot_addr, dtype=np.int64)
             # WQE ring: [0] = region offset, [1] = staging address
             self._desc_pin = torch.empty((2, self.inflight),
                                          dtype=torch.int64, pin_memory=True)
@@ -94,6 +97,29 @@ class SdmaTransport(Transport):
                 dst.copy_(self.staging[slot], non_blocking=True)
             else:
                 self.staging[slot].copy_(dst, non_blocking=True)
+
+    def post_many(self, start: int, n: int) -> None:
+        """Vectorized WQE writes (numpy) — the posting loop itself must
+        not bound small-message rates (a verbs app posts in C; measured:
+        scalar python post() capped 4 KiB at ~10 GB/s)."""
+        if self.engine != "kernel":
+            return super().post_many(start, n)
+        import numpy as np
+
+        done = 0
+        while done < n:
+            if self._pending >= self.inflight:
+                self.flush()
+            k = self._pending
+            take = min(n - done, self.inflight - k)
+            idx = np.arange(start + done, start + done + take,
+                            dtype=np.int64)
+            self._desc_np[0, k:k + take] = \
+                (idx % self.msgs_per_region) * self.msg_bytes
+            self._desc_np[1, k:k + take] = \
+                self._slot_addr_np[idx % self.inflight]
+            self._pending = k + take
+            done += take
 
     def flush(self) -> None:
         if self.engine == "kernel":
