@@ -325,6 +325,81 @@ static inline u64 dma_get_max_seg_size(struct device *dev)
 	return dev->max_seg ? dev->max_seg : 0x100000000ULL;
 }
 
+/* ---- char device / uaccess / mmap (for the probe module) ---- */
+struct inode {
+	int i_dummy;
+};
+struct file {
+	void *private_data;
+};
+struct vm_area_struct {
+	unsigned long vm_start;
+	unsigned long vm_end;
+	unsigned long vm_pgoff;
+	unsigned long vm_flags;
+	unsigned long vm_page_prot;
+};
+#define VM_READ		0x1UL
+#define VM_WRITE	0x2UL
+#define VM_SHARED	0x8UL
+#define VM_IO		0x4000UL
+#define VM_PFNMAP	0x400UL
+#define VM_DONTEXPAND	0x40000UL
+#define VM_DONTDUMP	0x4000000UL
+static inline void vm_flags_set(struct vm_area_struct *vma,
+				unsigned long flags)
+{
+	vma->vm_flags |= flags;
+}
+#define pgprot_noncached(p) (p)
+
+struct file_operations {
+	struct module *owner;
+	int (*open)(struct inode *, struct file *);
+	int (*release)(struct inode *, struct file *);
+	long (*unlocked_ioctl)(struct file *, unsigned int, unsigned long);
+	int (*mmap)(struct file *, struct vm_area_struct *);
+};
+
+struct miscdevice {
+	int minor;
+	const char *name;
+	const struct file_operations *fops;
+	unsigned short mode;
+};
+#define MISC_DYNAMIC_MINOR 255
+int misc_register(struct miscdevice *dev);
+void misc_deregister(struct miscdevice *dev);
+const struct miscdevice *rocnr_shim_misc_dev(void);
+
+static inline unsigned long copy_from_user(void *to, const void *from,
+					   unsigned long n)
+{
+	memcpy(to, from, n);
+	return 0;
+}
+static inline unsigned long copy_to_user(void *to, const void *from,
+					 unsigned long n)
+{
+	memcpy(to, from, n);
+	return 0;
+}
+
+/* io_remap_pfn_range recorder: tests assert the exact mappings the
+ * probe's mmap produced. */
+struct rocnr_shim_map {
+	unsigned long vaddr;
+	unsigned long pfn;
+	unsigned long size;
+};
+int io_remap_pfn_range(struct vm_area_struct *vma, unsigned long vaddr,
+		       unsigned long pfn, unsigned long size,
+		       unsigned long prot);
+#define remap_pfn_range io_remap_pfn_range
+void rocnr_shim_maps_reset(void);
+long rocnr_shim_maps_count(void);
+const struct rocnr_shim_map *rocnr_shim_maps_get(long i);
+
 #ifdef __cplusplus
 }
 #endif

@@ -138,3 +138,52 @@ void dma_unmap_resource(struct device *dev, dma_addr_t addr, size_t size,
 	(void)attrs;
 	atomic64_dec(&dev->live_maps);
 }
+
+/* ---- misc device ---- */
+static struct miscdevice *shim_misc;
+
+int misc_register(struct miscdevice *dev)
+{
+	if (shim_misc)
+		return -EBUSY;
+	shim_misc = dev;
+	return 0;
+}
+
+void misc_deregister(struct miscdevice *dev)
+{
+	if (shim_misc == dev)
+		shim_misc = NULL;
+}
+
+const struct miscdevice *rocnr_shim_misc_dev(void)
+{
+	return shim_misc;
+}
+
+/* ---- io_remap_pfn_range recorder ---- */
+#define SHIM_MAX_MAPS 512
+static struct rocnr_shim_map shim_maps[SHIM_MAX_MAPS];
+static long shim_nmaps;
+
+int io_remap_pfn_range(struct vm_area_struct *vma, unsigned long vaddr,
+		       unsigned long pfn, unsigned long size,
+		       unsigned long prot)
+{
+	(void)vma;
+	(void)prot;
+	if (shim_nmaps >= SHIM_MAX_MAPS)
+		return -ENOMEM;
+	shim_maps[shim_nmaps].vaddr = vaddr;
+	shim_maps[shim_nmaps].pfn = pfn;
+	shim_maps[shim_nmaps].size = size;
+	shim_nmaps++;
+	return 0;
+}
+
+void rocnr_shim_maps_reset(void) { shim_nmaps = 0; }
+long rocnr_shim_maps_count(void) { return shim_nmaps; }
+const struct rocnr_shim_map *rocnr_shim_maps_get(long i)
+{
+	return (i >= 0 && i < shim_nmaps) ? &shim_maps[i] : 0;
+}

@@ -32,3 +32,38 @@ def test_bridge_shim_suite_asan():
     out = subprocess.run([os.path.join(SHIM, "build", "bridge_tests_asan")],
                          capture_output=True, text=True, timeout=600)
     assert out.returncode == 0, out.stdout + out.stderr
+
+
+@pytest.mark.skipif(shutil.which("gcc") is None or
+                    shutil.which("make") is None, reason="no toolchain")
+@pytest.mark.timeout(600)
+def test_probe_shim_suite():
+    subprocess.run(["make", "-C", SHIM, "all"], check=True,
+                   capture_output=True, text=True)
+    out = subprocess.run([os.path.join(SHIM, "build", "probe_tests")],
+                         capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stdout + out.stderr
+    assert "ALL PROBE TESTS PASSED" in out.stdout
+
+
+@pytest.mark.skipif(shutil.which("gcc") is None or
+                    shutil.which("make") is None, reason="no toolchain")
+@pytest.mark.timeout(900)
+def test_probe_shim_suite_asan():
+    subprocess.run(["make", "-C", SHIM, "all"], check=True,
+                   capture_output=True, text=True)
+    out = subprocess.run([os.path.join(SHIM, "build", "probe_tests_asan")],
+                         capture_output=True, text=True, timeout=600)
+    assert out.returncode == 0, out.stdout + out.stderr
+
+
+@pytest.mark.skipif(shutil.which("gcc") is None, reason="no gcc")
+def test_probe_cli_compiles(tmp_path):
+    """The userspace ioctl client builds against the vendored ABI."""
+    out = subprocess.run(
+        ["gcc", "-O2", "-Wall", "-Werror",
+         os.path.join(ROOT, "tools", "rocp2p_probe_cli.c"),
+         "-I", os.path.join(ROOT, "module", "include"),
+         "-o", str(tmp_path / "cli")],
+        capture_output=True, text=True)
+    assert out.returncode == 0, out.stderr
