@@ -26,25 +26,13 @@
 #include <hip/hip_runtime.h>
 #include <cstdint>
 #include "p2p_kernels.h"
+#include "p2p_pattern.h"
 
 #define WAVE 64u
 #define PAGE 4096u
 #define SEG 64u  // bytes per lane within a page
 
-// ---------------------------------------------------------------------
-// splitmix64 pattern (reference implementation also in Python/numpy:
-// rocnrdma_amd/utils/pattern.py — must stay bit-identical)
-__host__ __device__ __forceinline__ uint64_t sm64_mix(uint64_t x) {
-  x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ULL;
-  x = (x ^ (x >> 27)) * 0x94D049BB133111EBULL;
-  return x ^ (x >> 31);
-}
-#define SM64_GOLDEN 0x9E3779B97F4A7C15ULL
-
-__host__ __device__ __forceinline__ uint64_t pattern_word(uint64_t seed,
-                                                          uint64_t i) {
-  return sm64_mix(seed + (i + 1) * SM64_GOLDEN);
-}
+#define pattern_word rocp2p_pattern_word
 
 // ---------------------------------------------------------------------
 // fill: 2 words (16 B) per lane per grid-stride step

@@ -1,0 +1,68 @@
+"""Builds and exercises the native C++ harness (rocp2p_bw) on the fake
+backend — CPU tier; its hip/verbs backends run on GPU/HCA hosts."""
+import json
+import os
+import shutil
+import subprocess
+
+import pytest
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+HARNESS = os.path.join(ROOT, "harness")
+BIN = os.path.join(HARNESS, "build", "rocp2p_bw")
+
+
+@pytest.fixture(scope="module")
+def built():
+    if shutil.which("make") is None or shutil.which("hipcc") is None:
+        pytest.skip("no toolchain")
+    subprocess.run(["make", "-C", HARNESS], check=True, capture_output=True,
+                   text=True, timeout=900)
+    return BIN
+
+
+@pytest.mark.timeout(900)
+def test_fake_write_json(built):
+    out = subprocess.run(
+        [built, "--transport", "fake", "--msg", "65536", "--region",
+         "4194304", "--secs", "0.2", "--json"],
+        capture_output=True, text=True, timeout=120)
+    assert out.returncode == 0, out.stderr
+    r = json.loads(out.stdout.strip())
+    assert r["transport"] == "fake"
+    assert r["gbps"] > 0
+    assert r["integrity"] == "ok"
+
+
+@pytest.mark.timeout(300)
+def test_fake_read_direction(built):
+    out = subprocess.run(
+        [built, "--transport", "fake", "--msg", "4096", "--region",
+         "1048576", "--secs", "0.1", "--dir", "read", "--json"],
+        capture_output=True, text=True, timeout=120)
+    assert out.returncode == 0, out.stderr
+    r = json.loads(out.stdout.strip())
+    assert r["direction"] == "read"
+    assert r["integrity"] == "ok"
+
+
+@pytest.mark.timeout(300)
+def test_sweep_mode(built):
+    out = subprocess.run(
+        [built, "--transport", "fake", "--region", "8388608", "--secs",
+         "0.05", "--sweep", "--json"],
+        capture_output=True, text=True, timeout=240)
+    assert out.returncode == 0, out.stderr
+    lines = [json.loads(l) for l in out.stdout.strip().splitlines()]
+    assert len(lines) == 5  # 4K..64M
+    assert all(r["integrity"] == "ok" for r in lines)
+
+
+@pytest.mark.timeout(300)
+def test_verbs_unavailable_is_actionable(built):
+    out = subprocess.run(
+        [built, "--transport", "verbs", "--msg", "4096", "--region",
+         "1048576", "--secs", "0.05"],
+        capture_output=True, text=True, timeout=120)
+    assert out.returncode == 3
+    assert "verbs" in out.stderr.lower()
