@@ -73,9 +73,14 @@ def main():
 
     has_gpu = torch.cuda.is_available()
     device = None
+    numa = None
     if has_gpu:
         torch.cuda.set_device(local_rank % torch.cuda.device_count())
         device = torch.device("cuda", local_rank % torch.cuda.device_count())
+        # pin the posting loop + pinned staging to the GPU's NUMA node
+        from rocnrdma_amd.utils import topology
+
+        numa = topology.bind_rank_near_gpu(device.index)
 
     from rocnrdma_amd.transport import get_transport
 
@@ -164,6 +169,7 @@ def main():
             "inflight": args.inflight,
             "streams": args.streams,
             "integrity": integrity,
+            "numa_bound": numa,
         },
     }
     if rank == 0:

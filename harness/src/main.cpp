@@ -79,7 +79,7 @@ int main(int argc, char** argv) {
   TransportConfig cfg;
   double secs = 1.0;
   int gpus = 1;
-  bool sweep = false, json = false, integrity = true;
+  bool sweep = false, json = false, integrity = true, bidir = false;
   uint64_t seed = 0xC0FFEE;
 
   for (int i = 1; i < argc; i++) {
@@ -92,8 +92,11 @@ int main(int argc, char** argv) {
     else if (a == "--msg") cfg.msg_bytes = strtoull(next().c_str(), 0, 0);
     else if (a == "--region")
       cfg.region_bytes = strtoull(next().c_str(), 0, 0);
-    else if (a == "--dir")
-      cfg.dir = next() == "read" ? Direction::Read : Direction::Write;
+    else if (a == "--dir") {
+      std::string d = next();
+      if (d == "bidir") bidir = true;
+      cfg.dir = d == "read" ? Direction::Read : Direction::Write;
+    }
     else if (a == "--secs") secs = atof(next().c_str());
     else if (a == "--gpus") gpus = atoi(next().c_str());
     else if (a == "--engine") cfg.engine = next();
@@ -124,20 +127,25 @@ int main(int argc, char** argv) {
     c.msg_bytes = msg;
     c.region_bytes = std::max(cfg.region_bytes / msg, (size_t)1) * msg;
 
-    std::vector<Result> res(gpus);
-    std::vector<std::string> names(gpus);
+    // --dir bidir: two workers per GPU (write + read concurrently) —
+    // full-duplex PCIe, the ib_write_bw --bidirectional analog
+    int workers = bidir ? 2 * gpus : gpus;
+    std::vector<Result> res(workers);
+    std::vector<std::string> names(workers);
     std::vector<std::thread> ths;
-    std::vector<std::string> errs(gpus);
-    for (int g = 0; g < gpus; g++) {
-      ths.emplace_back([&, g] {
+    std::vector<std::string> errs(workers);
+    for (int w = 0; w < workers; w++) {
+      ths.emplace_back([&, w] {
         try {
           TransportConfig cg = c;
-          cg.device_index = g;
+          cg.device_index = bidir ? w / 2 : w;
+          if (bidir)
+            cg.dir = (w & 1) ? Direction::Read : Direction::Write;
           auto tp = make_transport(transport, cg);
-          names[g] = tp->name();
-          res[g] = run_point(*tp, secs, seed + g, integrity);
+          names[w] = tp->name();
+          res[w] = run_point(*tp, secs, seed + w, integrity);
         } catch (const std::exception& e) {
-          errs[g] = e.what();
+          errs[w] = e.what();
         }
       });
     }
@@ -145,14 +153,14 @@ int main(int argc, char** argv) {
 
     double agg = 0;
     uint64_t bad = 0, msgs = 0;
-    for (int g = 0; g < gpus; g++) {
-      if (!errs[g].empty()) {
-        fprintf(stderr, "gpu %d: %s\n", g, errs[g].c_str());
+    for (int w = 0; w < workers; w++) {
+      if (!errs[w].empty()) {
+        fprintf(stderr, "worker %d: %s\n", w, errs[w].c_str());
         return 3;
       }
-      agg += res[g].gbps;
-      msgs += res[g].msgs;
-      if (integrity) bad += res[g].bad;
+      agg += res[w].gbps;
+      msgs += res[w].msgs;
+      if (integrity) bad += res[w].bad;
     }
     const char* ok = !integrity ? "skipped" : (bad == 0 ? "ok" : "FAILED");
     if (json) {
@@ -160,12 +168,12 @@ int main(int argc, char** argv) {
              "\"gpus\":%d,\"gbps\":%.3f,\"msgs_per_s\":%.0f,"
              "\"integrity\":\"%s\"}\n",
              names[0].c_str(), msg,
-             c.dir == Direction::Write ? "write" : "read", gpus, agg,
-             msgs / res[0].secs, ok);
+             bidir ? "bidir" : (c.dir == Direction::Write ? "write" : "read"),
+             gpus, agg, msgs / res[0].secs, ok);
     } else {
       printf("%12zu %6s %4d %10.3f %12.0f %10s\n", msg,
-             c.dir == Direction::Write ? "write" : "read", gpus, agg,
-             msgs / res[0].secs, ok);
+             bidir ? "bidir" : (c.dir == Direction::Write ? "write" : "read"),
+             gpus, agg, msgs / res[0].secs, ok);
     }
     if (integrity && bad) rc = 1;
   }
