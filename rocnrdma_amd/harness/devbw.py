@@ -36,6 +36,8 @@ def bench_kernels(nbytes: int, iters: int = 10) -> dict:
     out["verify_GBps"] = timeit(lambda: ops.verify(buf, 1), nbytes)
     out["crc32_GBps"] = timeit(lambda: ops.crc32_pages(buf), nbytes)
     out["copy_GBps_rw"] = timeit(lambda: ops.copy_(dst, buf), 2 * nbytes)
+    out["copy_nt_GBps_rw"] = timeit(lambda: ops.copy_nt_(dst, buf),
+                                    2 * nbytes)
     return out
 
 

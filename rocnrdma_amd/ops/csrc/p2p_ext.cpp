@@ -101,6 +101,15 @@ void copy_(torch::Tensor dst, torch::Tensor src) {
                      stream.stream()));
 }
 
+void copy_nt_(torch::Tensor dst, torch::Tensor src) {
+  check_buf(dst, "copy_nt_");
+  check_buf(src, "copy_nt_");
+  TORCH_CHECK(nbytes_of(dst) == nbytes_of(src), "copy_nt_: size mismatch");
+  auto stream = at::cuda::getCurrentCUDAStream();
+  HIP_OK(rocp2p_copy_nt(dst.data_ptr(), src.data_ptr(), nbytes_of(dst),
+                        stream.stream()));
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -110,6 +119,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("verify", &verify, "count words deviating from the pattern");
   m.def("crc32_pages", &crc32_pages, "zlib CRC32 of each 4 KiB page");
   m.def("copy_", &copy_, "streaming device copy dst <- src");
+  m.def("copy_nt_", &copy_nt_, "nontemporal streaming device copy");
   m.def("gather_", &gather_,
         "batched message engine: host-pinned srcs -> HBM region offsets");
   m.def("scatter_", &scatter_,

@@ -30,6 +30,9 @@ hipError_t rocp2p_crc32_pages(const void* buf, uint64_t npages,
 // Streaming device copy (bandwidth ceiling probe), 16 B/lane.
 hipError_t rocp2p_copy(void* dst, const void* src, uint64_t nbytes,
                        hipStream_t stream);
+// Nontemporal variant (streamed-once data; see microarch nt rows).
+hipError_t rocp2p_copy_nt(void* dst, const void* src, uint64_t nbytes,
+                          hipStream_t stream);
 
 // GPU-driven batched message engine (the NIC-WQE analog: one launch
 // retires a whole queue of posted messages).  src_addrs/dst_offs are

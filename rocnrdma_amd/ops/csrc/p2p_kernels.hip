@@ -81,6 +81,21 @@ __global__ void k_copy(uint4* __restrict__ dst, const uint4* __restrict__ src,
     dst[i] = src[i];
 }
 
+// nontemporal variant: streamed data read/written once should not
+// displace L2 residents (MI355X_MICROARCH.md nt-weights row).  The
+// builtin wants a clang vector, not HIP's uint4 class.
+typedef unsigned int uint4_cv __attribute__((ext_vector_type(4)));
+
+__global__ void k_copy_nt(uint4_cv* __restrict__ dst,
+                          const uint4_cv* __restrict__ src, uint64_t nvec) {
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (uint64_t i = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x;
+       i < nvec; i += stride) {
+    uint4_cv v = __builtin_nontemporal_load(&src[i]);
+    __builtin_nontemporal_store(v, &dst[i]);
+  }
+}
+
 // ---------------------------------------------------------------------
 // Batched message engine.  A real HCA retires posted WQEs with its DMA
 // hardware regardless of message size; the host-API (hipMemcpyAsync)
@@ -282,6 +297,16 @@ extern "C" hipError_t rocp2p_copy(void* dst, const void* src, uint64_t nbytes,
   uint32_t grid = stream_grid(nvec, 256);
   hipLaunchKernelGGL(k_copy, dim3(grid), dim3(256), 0, stream, (uint4*)dst,
                      (const uint4*)src, nvec);
+  return hipGetLastError();
+}
+
+extern "C" hipError_t rocp2p_copy_nt(void* dst, const void* src,
+                                     uint64_t nbytes, hipStream_t stream) {
+  if (nbytes % 16) return hipErrorInvalidValue;
+  uint64_t nvec = nbytes / 16;
+  uint32_t grid = stream_grid(nvec, 256);
+  hipLaunchKernelGGL(k_copy_nt, dim3(grid), dim3(256), 0, stream,
+                     (uint4_cv*)dst, (const uint4_cv*)src, nvec);
   return hipGetLastError();
 }
 
