@@ -48,6 +48,11 @@ def copy_(dst: torch.Tensor, src: torch.Tensor) -> None:
     _require().copy_(dst, src)
 
 
+def copy_nt_(dst: torch.Tensor, src: torch.Tensor) -> None:
+    """Nontemporal streaming copy (streamed-once data stays out of L2)."""
+    _require().copy_nt_(dst, src)
+
+
 def gather_(region: torch.Tensor, dst_offs: torch.Tensor,
             src_addrs: torch.Tensor, msg_bytes: int) -> None:
     """Batched message engine: host-pinned sources -> HBM region offsets

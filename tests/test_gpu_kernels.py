@@ -81,6 +81,16 @@ def test_copy_kernel(dev):
     assert torch.equal(dst, src)
 
 
+def test_copy_nt_kernel(dev):
+    import rocnrdma_amd.ops as ops
+
+    src = torch.randint(0, 256, (8 << 20,), dtype=torch.uint8, device=dev)
+    dst = torch.zeros_like(src)
+    ops.copy_nt_(dst, src)
+    torch.cuda.synchronize()
+    assert torch.equal(dst, src)
+
+
 @pytest.mark.parametrize("engine", ["stream", "kernel"])
 @pytest.mark.parametrize("direction", ["write", "read"])
 def test_sdma_transport_integrity(dev, direction, engine):
