@@ -35,18 +35,42 @@
 #define pattern_word rocp2p_pattern_word
 
 // ---------------------------------------------------------------------
-// fill: 2 words (16 B) per lane per grid-stride step
+// Streaming-kernel shape (tuned on MI355X, tools/hbm_bench.hip): each
+// WAVE owns a contiguous tile of UNROLL x 1 KiB; a lane's UNROLL
+// accesses are 64 vectors apart so every instruction stays a fully
+// coalesced 1 KiB wave access, with UNROLL independent loads/stores in
+// flight per lane.  Nontemporal policy on streamed-once data.
+// Measured vs the naive 1-vec grid-stride loop: fill 4.0 -> 5.6 TB/s,
+// copy 4.7 -> 5.7 TB/s (r+w).
+typedef unsigned long long u64v2_cv __attribute__((ext_vector_type(2)));
+
+// fill: UNROLL=8, nt stores
 __global__ void k_fill(uint64_t* __restrict__ buf, uint64_t nwords,
                        uint64_t seed) {
-  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
-  for (uint64_t i = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x;
-       i * 2 + 1 < nwords; i += stride) {
-    uint64_t w0 = pattern_word(seed, i * 2);
-    uint64_t w1 = pattern_word(seed, i * 2 + 1);
-    // single dwordx4 store
-    reinterpret_cast<ulonglong2*>(buf)[i] = make_ulonglong2(w0, w1);
+  constexpr int U = 8;
+  u64v2_cv* out = reinterpret_cast<u64v2_cv*>(buf);
+  uint64_t nvec = nwords / 2;
+  uint64_t nthreads = (uint64_t)gridDim.x * blockDim.x;
+  uint64_t tid = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x;
+  uint64_t nwaves = nthreads / WAVE;
+  uint64_t wave = tid / WAVE, lane = tid & (WAVE - 1);
+  const uint64_t tile = (uint64_t)WAVE * U;
+  uint64_t ntiles = nvec / tile;
+  for (uint64_t t = wave; t < ntiles; t += nwaves) {
+    uint64_t base = t * tile + lane;
+#pragma unroll
+    for (int u = 0; u < U; u++) {
+      uint64_t j = base + (uint64_t)WAVE * u;
+      u64v2_cv v = {pattern_word(seed, j * 2), pattern_word(seed, j * 2 + 1)};
+      __builtin_nontemporal_store(v, &out[j]);
+    }
   }
-  // odd tail word
+  // vector tail
+  for (uint64_t j = ntiles * tile + tid; j < nvec; j += nthreads) {
+    u64v2_cv v = {pattern_word(seed, j * 2), pattern_word(seed, j * 2 + 1)};
+    __builtin_nontemporal_store(v, &out[j]);
+  }
+  // odd word tail
   if (blockIdx.x == 0 && threadIdx.x == 0 && (nwords & 1))
     buf[nwords - 1] = pattern_word(seed, nwords - 1);
 }
@@ -72,28 +96,37 @@ __global__ void k_verify(const uint64_t* __restrict__ buf, uint64_t nwords,
 }
 
 // ---------------------------------------------------------------------
-// streaming copy, 16 B/lane
-__global__ void k_copy(uint4* __restrict__ dst, const uint4* __restrict__ src,
-                       uint64_t nvec) {
-  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
-  for (uint64_t i = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x;
-       i < nvec; i += stride)
-    dst[i] = src[i];
-}
-
-// nontemporal variant: streamed data read/written once should not
-// displace L2 residents (MI355X_MICROARCH.md nt-weights row).  The
-// builtin wants a clang vector, not HIP's uint4 class.
+// streaming copy (same wave-tile shape; UNROLL=4).  The builtin wants a
+// clang vector, not HIP's uint4 class.
 typedef unsigned int uint4_cv __attribute__((ext_vector_type(4)));
 
-__global__ void k_copy_nt(uint4_cv* __restrict__ dst,
-                          const uint4_cv* __restrict__ src, uint64_t nvec) {
-  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
-  for (uint64_t i = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x;
-       i < nvec; i += stride) {
-    uint4_cv v = __builtin_nontemporal_load(&src[i]);
-    __builtin_nontemporal_store(v, &dst[i]);
+template <bool NT>
+__global__ void k_copy_t(uint4_cv* __restrict__ dst,
+                         const uint4_cv* __restrict__ src, uint64_t nvec) {
+  constexpr int U = 4;
+  uint64_t nthreads = (uint64_t)gridDim.x * blockDim.x;
+  uint64_t tid = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x;
+  uint64_t nwaves = nthreads / WAVE;
+  uint64_t wave = tid / WAVE, lane = tid & (WAVE - 1);
+  const uint64_t tile = (uint64_t)WAVE * U;
+  uint64_t ntiles = nvec / tile;
+  for (uint64_t t = wave; t < ntiles; t += nwaves) {
+    uint64_t base = t * tile + lane;
+    uint4_cv v[U];
+#pragma unroll
+    for (int u = 0; u < U; u++)
+      v[u] = NT ? __builtin_nontemporal_load(&src[base + WAVE * u])
+                : src[base + WAVE * u];
+#pragma unroll
+    for (int u = 0; u < U; u++) {
+      if (NT)
+        __builtin_nontemporal_store(v[u], &dst[base + WAVE * u]);
+      else
+        dst[base + WAVE * u] = v[u];
+    }
   }
+  for (uint64_t i = ntiles * tile + tid; i < nvec; i += nthreads)
+    dst[i] = src[i];
 }
 
 // ---------------------------------------------------------------------
@@ -263,7 +296,7 @@ __global__ void __launch_bounds__(256) k_crc32_pages(
 
 static inline uint32_t stream_grid(uint64_t items, uint32_t per_block) {
   uint64_t blocks = (items + per_block - 1) / per_block;
-  if (blocks > 4096) blocks = 4096;  // 16 WGs/CU worth of grid-stride
+  if (blocks > 16384) blocks = 16384;  // tuned: tools/hbm_bench.hip
   if (blocks == 0) blocks = 1;
   return (uint32_t)blocks;
 }
@@ -272,7 +305,7 @@ extern "C" hipError_t rocp2p_fill(void* buf, uint64_t nbytes, uint64_t seed,
                                   hipStream_t stream) {
   if (nbytes % 8) return hipErrorInvalidValue;
   uint64_t nwords = nbytes / 8;
-  uint32_t grid = stream_grid(nwords / 2, 256);
+  uint32_t grid = stream_grid(nwords / 16, 256);
   hipLaunchKernelGGL(k_fill, dim3(grid), dim3(256), 0, stream,
                      (uint64_t*)buf, nwords, seed);
   return hipGetLastError();
@@ -294,9 +327,9 @@ extern "C" hipError_t rocp2p_copy(void* dst, const void* src, uint64_t nbytes,
                                   hipStream_t stream) {
   if (nbytes % 16) return hipErrorInvalidValue;
   uint64_t nvec = nbytes / 16;
-  uint32_t grid = stream_grid(nvec, 256);
-  hipLaunchKernelGGL(k_copy, dim3(grid), dim3(256), 0, stream, (uint4*)dst,
-                     (const uint4*)src, nvec);
+  uint32_t grid = stream_grid(nvec / 4, 256);
+  hipLaunchKernelGGL(k_copy_t<false>, dim3(grid), dim3(256), 0, stream,
+                     (uint4_cv*)dst, (const uint4_cv*)src, nvec);
   return hipGetLastError();
 }
 
@@ -304,8 +337,8 @@ extern "C" hipError_t rocp2p_copy_nt(void* dst, const void* src,
                                      uint64_t nbytes, hipStream_t stream) {
   if (nbytes % 16) return hipErrorInvalidValue;
   uint64_t nvec = nbytes / 16;
-  uint32_t grid = stream_grid(nvec, 256);
-  hipLaunchKernelGGL(k_copy_nt, dim3(grid), dim3(256), 0, stream,
+  uint32_t grid = stream_grid(nvec / 4, 256);
+  hipLaunchKernelGGL(k_copy_t<true>, dim3(grid), dim3(256), 0, stream,
                      (uint4_cv*)dst, (const uint4_cv*)src, nvec);
   return hipGetLastError();
 }
