@@ -10,6 +10,7 @@
 //
 // --gpus N fans out one transport ("QP") per GPU on N worker threads and
 // reports per-GPU + aggregate bandwidth (BASELINE configs 4-5).
+#include <algorithm>
 #include <chrono>
 #include <cstdio>
 #include <cstdlib>
@@ -46,7 +47,34 @@ using clk = std::chrono::steady_clock;
 struct Result {
   double gbps = 0, secs = 0;
   uint64_t msgs = 0, bad = ~0ull;
+  double lat_min = 0, lat_p50 = 0, lat_p99 = 0, lat_max = 0;  // us
 };
+
+// ib_write_lat analog: one message at a time, completion-to-completion
+static Result run_lat(Transport& tp, int iters, uint64_t seed,
+                      bool integrity) {
+  Result r;
+  tp.post_many(0, 1);
+  tp.flush();
+  std::vector<double> us;
+  us.reserve(iters);
+  for (int i = 0; i < iters; i++) {
+    auto t0 = clk::now();
+    tp.post_many(i, 1);
+    tp.flush();
+    us.push_back(
+        std::chrono::duration<double, std::micro>(clk::now() - t0).count());
+  }
+  std::sort(us.begin(), us.end());
+  r.msgs = iters;
+  r.lat_min = us.front();
+  r.lat_p50 = us[us.size() / 2];
+  r.lat_p99 = us[(size_t)(us.size() * 0.99)];
+  r.lat_max = us.back();
+  r.secs = 0;
+  if (integrity) r.bad = tp.integrity_check(seed);
+  return r;
+}
 
 static Result run_point(Transport& tp, double secs, uint64_t seed,
                         bool integrity) {
@@ -80,6 +108,7 @@ int main(int argc, char** argv) {
   double secs = 1.0;
   int gpus = 1;
   bool sweep = false, json = false, integrity = true, bidir = false;
+  int lat_iters = 0;
   uint64_t seed = 0xC0FFEE;
 
   for (int i = 1; i < argc; i++) {
@@ -104,6 +133,7 @@ int main(int argc, char** argv) {
     else if (a == "--streams") cfg.num_streams = atoi(next().c_str());
     else if (a == "--inflight") cfg.inflight = strtoull(next().c_str(), 0, 0);
     else if (a == "--seed") seed = strtoull(next().c_str(), 0, 0);
+    else if (a == "--lat") lat_iters = atoi(next().c_str());
     else if (a == "--sweep") sweep = true;
     else if (a == "--json") json = true;
     else if (a == "--no-integrity") integrity = false;
@@ -143,7 +173,9 @@ int main(int argc, char** argv) {
             cg.dir = (w & 1) ? Direction::Read : Direction::Write;
           auto tp = make_transport(transport, cg);
           names[w] = tp->name();
-          res[w] = run_point(*tp, secs, seed + w, integrity);
+          res[w] = lat_iters
+                       ? run_lat(*tp, lat_iters, seed + w, integrity)
+                       : run_point(*tp, secs, seed + w, integrity);
         } catch (const std::exception& e) {
           errs[w] = e.what();
         }
@@ -163,6 +195,21 @@ int main(int argc, char** argv) {
       if (integrity) bad += res[w].bad;
     }
     const char* ok = !integrity ? "skipped" : (bad == 0 ? "ok" : "FAILED");
+    if (lat_iters) {
+      // latency lines (per GPU 0; multi-gpu latency is per-worker)
+      if (json)
+        printf("{\"transport\":\"%s\",\"msg_bytes\":%zu,\"mode\":\"lat\","
+               "\"iters\":%d,\"us_min\":%.2f,\"us_p50\":%.2f,"
+               "\"us_p99\":%.2f,\"us_max\":%.2f,\"integrity\":\"%s\"}\n",
+               names[0].c_str(), msg, lat_iters, res[0].lat_min,
+               res[0].lat_p50, res[0].lat_p99, res[0].lat_max, ok);
+      else
+        printf("%12zu lat us: min %.2f p50 %.2f p99 %.2f max %.2f  %s\n",
+               msg, res[0].lat_min, res[0].lat_p50, res[0].lat_p99,
+               res[0].lat_max, ok);
+      if (integrity && bad) rc = 1;
+      continue;
+    }
     if (json) {
       printf("{\"transport\":\"%s\",\"msg_bytes\":%zu,\"direction\":\"%s\","
              "\"gpus\":%d,\"gbps\":%.3f,\"msgs_per_s\":%.0f,"

@@ -50,10 +50,36 @@ def run_point(tp, target_secs: float = 1.0, min_msgs: int = 8) -> dict:
     }
 
 
+def run_lat_point(tp, iters: int = 1000) -> dict:
+    """ib_write_lat analog: single-message completion latency."""
+    import time as _t
+
+    tp.post_many(0, 1)
+    tp.flush()
+    us = []
+    for i in range(iters):
+        t0 = _t.perf_counter()
+        tp.post_many(i, 1)
+        tp.flush()
+        us.append((_t.perf_counter() - t0) * 1e6)
+    us.sort()
+    return {
+        "msg_bytes": tp.msg_bytes,
+        "direction": tp.direction,
+        "mode": "lat",
+        "iters": iters,
+        "us_min": round(us[0], 2),
+        "us_p50": round(us[len(us) // 2], 2),
+        "us_p99": round(us[int(len(us) * 0.99)], 2),
+        "us_max": round(us[-1], 2),
+    }
+
+
 def run_sweep(transport: str = "auto", region_bytes: int = 1 << 30,
               sizes=None, directions=("write", "read"),
               target_secs: float = 1.0, device=None,
-              num_streams: int = 2, inflight: int | None = 0) -> list[dict]:
+              num_streams: int = 2, inflight: int | None = 0,
+              lat_iters: int = 0) -> list[dict]:
     from rocnrdma_amd.transport import get_transport
 
     rows = []
@@ -65,7 +91,10 @@ def run_sweep(transport: str = "auto", region_bytes: int = 1 << 30,
                                device=device, num_streams=num_streams,
                                inflight=inflight)
             try:
-                row = run_point(tp, target_secs=target_secs)
+                if lat_iters:
+                    row = run_lat_point(tp, lat_iters)
+                else:
+                    row = run_point(tp, target_secs=target_secs)
                 row["transport"] = tp.name
                 rows.append(row)
             finally:
@@ -82,6 +111,8 @@ def main():
     ap.add_argument("--inflight", type=int, default=0)
     ap.add_argument("--sizes", default="")
     ap.add_argument("--out", default="")
+    ap.add_argument("--lat", type=int, default=0,
+                    help="latency mode: N single-message iterations")
     args = ap.parse_args()
 
     sizes = ([int(s) for s in args.sizes.split(",")] if args.sizes
@@ -92,12 +123,19 @@ def main():
         args.region_bytes = min(args.region_bytes, 256 << 20)
     rows = run_sweep(args.transport, args.region_bytes, sizes,
                      target_secs=args.secs, num_streams=args.streams,
-                     inflight=args.inflight)
-    hdr = f"{'msg':>12} {'dir':>6} {'GB/s':>10} {'Mmsg/s':>10}"
-    print(hdr)
-    for r in rows:
-        print(f"{r['msg_bytes']:>12} {r['direction']:>6} "
-              f"{r['gbps']:>10.3f} {r['mops']:>10.4f}")
+                     inflight=args.inflight, lat_iters=args.lat)
+    if args.lat:
+        print(f"{'msg':>12} {'dir':>6} {'us_min':>8} {'us_p50':>8} "
+              f"{'us_p99':>8} {'us_max':>8}")
+        for r in rows:
+            print(f"{r['msg_bytes']:>12} {r['direction']:>6} "
+                  f"{r['us_min']:>8.2f} {r['us_p50']:>8.2f} "
+                  f"{r['us_p99']:>8.2f} {r['us_max']:>8.2f}")
+    else:
+        print(f"{'msg':>12} {'dir':>6} {'GB/s':>10} {'Mmsg/s':>10}")
+        for r in rows:
+            print(f"{r['msg_bytes']:>12} {r['direction']:>6} "
+                  f"{r['gbps']:>10.3f} {r['mops']:>10.4f}")
     if args.out:
         with open(args.out, "w") as f:
             json.dump(rows, f, indent=1)

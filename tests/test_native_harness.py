@@ -66,3 +66,16 @@ def test_verbs_unavailable_is_actionable(built):
         capture_output=True, text=True, timeout=120)
     assert out.returncode == 3
     assert "verbs" in out.stderr.lower()
+
+
+@pytest.mark.timeout(300)
+def test_lat_mode(built):
+    out = subprocess.run(
+        [built, "--transport", "fake", "--msg", "4096", "--region",
+         "1048576", "--lat", "500", "--json"],
+        capture_output=True, text=True, timeout=120)
+    assert out.returncode == 0, out.stderr
+    r = json.loads(out.stdout.strip())
+    assert r["mode"] == "lat"
+    assert 0 < r["us_min"] <= r["us_p50"] <= r["us_p99"] <= r["us_max"]
+    assert r["integrity"] == "ok"

@@ -54,3 +54,12 @@ def test_inflight_clamped():
     tp = get_transport("fake", msg_bytes=4096, region_bytes=8192,
                        inflight=64)
     assert tp.inflight == 2
+
+
+def test_python_lat_sweep():
+    from rocnrdma_amd.harness.sweep import run_lat_point
+
+    tp = get_transport("fake", msg_bytes=4096, region_bytes=65536)
+    r = run_lat_point(tp, iters=200)
+    assert r["mode"] == "lat"
+    assert 0 < r["us_min"] <= r["us_p50"] <= r["us_p99"] <= r["us_max"]
