@@ -64,3 +64,9 @@ def scatter_(region: torch.Tensor, src_offs: torch.Tensor,
              dst_addrs: torch.Tensor, msg_bytes: int) -> None:
     """Batched message engine: HBM region offsets -> host-pinned dests."""
     _require().scatter_(region, src_offs, dst_addrs, msg_bytes)
+
+
+def dmabuf_fd(buf: torch.Tensor) -> int:
+    """Export an HBM tensor as a dmabuf fd (ibv_reg_dmabuf_mr's GPU
+    half). Caller must os.close() the fd."""
+    return _require().dmabuf_fd(buf)

@@ -110,6 +110,21 @@ void copy_nt_(torch::Tensor dst, torch::Tensor src) {
                         stream.stream()));
 }
 
+// Export an HBM range as a dmabuf fd — the GPU half of the verbs
+// backend's kernel-module-free MR mode (ibv_reg_dmabuf_mr); lets a
+// GPU-only box validate that machinery without an HCA.  Caller owns
+// the fd (os.close it).
+int64_t dmabuf_fd(torch::Tensor buf) {
+  check_buf(buf, "dmabuf_fd");
+  int fd = -1;
+  hipError_t e = hipMemGetHandleForAddressRange(
+      &fd, buf.data_ptr(), nbytes_of(buf), hipMemRangeHandleTypeDmaBufFd,
+      0);
+  TORCH_CHECK(e == hipSuccess, "hipMemGetHandleForAddressRange: ",
+              hipGetErrorString(e));
+  return fd;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -124,4 +139,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "batched message engine: host-pinned srcs -> HBM region offsets");
   m.def("scatter_", &scatter_,
         "batched message engine: HBM region offsets -> host-pinned dsts");
+  m.def("dmabuf_fd", &dmabuf_fd,
+        "export an HBM tensor range as a dmabuf fd (caller closes)");
 }

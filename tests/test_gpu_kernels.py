@@ -141,3 +141,18 @@ def test_smoke_entry():
     import __graft_entry__ as ge
 
     ge.smoke()
+
+
+def test_dmabuf_export(dev):
+    """The kernel-module-free MR path's GPU half: an HBM range exports
+    as a dmabuf fd (what ibv_reg_dmabuf_mr consumes on HCA hosts)."""
+    import os
+
+    import rocnrdma_amd.ops as ops
+
+    buf = torch.empty(8 << 20, dtype=torch.uint8, device=dev)
+    fd = ops.dmabuf_fd(buf)
+    assert fd >= 0
+    st = os.fstat(fd)
+    assert st is not None
+    os.close(fd)
