@@ -23,6 +23,7 @@ struct TransportConfig {
   size_t inflight = 0;       // 0 = backend-chosen
   std::string engine = "auto";  // hip: auto|kernel|stream
   std::string verbs_mr = "auto";  // verbs: auto|peer|dmabuf|host
+  bool wc_staging = false;   // hip: write-combined pinned staging
 };
 
 class Transport {

@@ -59,7 +59,12 @@ class HipTransport final : public Transport {
 
     streams_.resize(std::max(1, cfg.num_streams));
     for (auto& s : streams_) HIP_THROW(hipStreamCreate(&s));
-    HIP_THROW(hipHostMalloc(&staging_, inflight_ * cfg.msg_bytes, 0));
+    // write-combined staging: uncached host stores, faster device reads
+    // over PCIe for the write direction (host-side verify in read/
+    // integrity paths gets slower uncached loads — measure both)
+    HIP_THROW(hipHostMalloc(&staging_, inflight_ * cfg.msg_bytes,
+                            cfg.wc_staging ? hipHostMallocWriteCombined
+                                           : 0u));
     HIP_THROW(hipMalloc(&region_, cfg.region_bytes));
     HIP_THROW(hipMemsetAsync(region_, 0, cfg.region_bytes, streams_[0]));
     HIP_THROW(hipMalloc(&d_mismatch_, sizeof(unsigned long long)));

@@ -133,6 +133,7 @@ int main(int argc, char** argv) {
     else if (a == "--streams") cfg.num_streams = atoi(next().c_str());
     else if (a == "--inflight") cfg.inflight = strtoull(next().c_str(), 0, 0);
     else if (a == "--seed") seed = strtoull(next().c_str(), 0, 0);
+    else if (a == "--wc") cfg.wc_staging = true;
     else if (a == "--lat") lat_iters = atoi(next().c_str());
     else if (a == "--sweep") sweep = true;
     else if (a == "--json") json = true;
