@@ -78,13 +78,33 @@ __global__ void k_fill(uint64_t* __restrict__ buf, uint64_t nwords,
 __global__ void k_verify(const uint64_t* __restrict__ buf, uint64_t nwords,
                          uint64_t seed,
                          unsigned long long* __restrict__ mismatch) {
-  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  constexpr int U = 4;
+  const u64v2_cv* in = reinterpret_cast<const u64v2_cv*>(buf);
+  uint64_t nvec = nwords / 2;
+  uint64_t nthreads = (uint64_t)gridDim.x * blockDim.x;
+  uint64_t tid = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x;
+  uint64_t nwaves = nthreads / WAVE;
+  uint64_t wave = tid / WAVE, lane = tid & (WAVE - 1);
+  const uint64_t tile = (uint64_t)WAVE * U;
+  uint64_t ntiles = nvec / tile;
   uint32_t bad = 0;
-  for (uint64_t i = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x;
-       i * 2 + 1 < nwords; i += stride) {
-    ulonglong2 v = reinterpret_cast<const ulonglong2*>(buf)[i];
-    bad += (v.x != pattern_word(seed, i * 2));
-    bad += (v.y != pattern_word(seed, i * 2 + 1));
+  for (uint64_t t = wave; t < ntiles; t += nwaves) {
+    uint64_t base = t * tile + lane;
+    u64v2_cv v[U];
+#pragma unroll
+    for (int u = 0; u < U; u++)
+      v[u] = __builtin_nontemporal_load(&in[base + WAVE * u]);
+#pragma unroll
+    for (int u = 0; u < U; u++) {
+      uint64_t j = base + (uint64_t)WAVE * u;
+      bad += (v[u].x != pattern_word(seed, j * 2));
+      bad += (v[u].y != pattern_word(seed, j * 2 + 1));
+    }
+  }
+  for (uint64_t j = ntiles * tile + tid; j < nvec; j += nthreads) {
+    u64v2_cv v = __builtin_nontemporal_load(&in[j]);
+    bad += (v.x != pattern_word(seed, j * 2));
+    bad += (v.y != pattern_word(seed, j * 2 + 1));
   }
   if (blockIdx.x == 0 && threadIdx.x == 0 && (nwords & 1))
     bad += (buf[nwords - 1] != pattern_word(seed, nwords - 1));
@@ -180,11 +200,15 @@ __global__ void k_scatter(const uint8_t* __restrict__ src_base,
 // CRC32 (zlib polynomial 0xEDB88320, init/xorout 0xFFFFFFFF)
 
 __constant__ uint32_t c_crc_tab[8][256];   // slice-by-8
-__constant__ uint32_t c_shift_mat[6][32];  // GF(2) ops: shift by 64<<k bytes
+// Byte-sliced GF(2) shift operators: c_shift_tab[k][b][v] = M(64B<<k)
+// applied to (v << 8b).  Applying a 32x32 GF(2) matrix becomes 4 table
+// lookups instead of 32 serial dependent steps.
+__constant__ uint32_t c_shift_tab[6][4][256];
 
 static bool g_crc_ready = false;
 
-static void host_make_tables(uint32_t tab[8][256], uint32_t mats[6][32]) {
+static void host_make_tables(uint32_t tab[8][256],
+                             uint32_t stab[6][4][256]) {
   const uint32_t POLY = 0xEDB88320u;
   for (uint32_t i = 0; i < 256; i++) {
     uint32_t c = i;
@@ -220,19 +244,12 @@ static void host_make_tables(uint32_t tab[8][256], uint32_t mats[6][32]) {
     for (int i = 0; i < 32; i++) cur[i] = m1[i];
   }  // cur = shift by 64 bytes
   for (int k = 0; k < 6; k++) {
-    for (int i = 0; i < 32; i++) mats[k][i] = cur[i];
+    for (int b = 0; b < 4; b++)
+      for (uint32_t v = 0; v < 256; v++)
+        stab[k][b][v] = times(cur, v << (8 * b));
     square(cur, m1);
     for (int i = 0; i < 32; i++) cur[i] = m1[i];
   }
-}
-
-__device__ __forceinline__ uint32_t lds_mat_times(const uint32_t* m,
-                                                  uint32_t v) {
-  uint32_t s = 0;
-#pragma unroll 1
-  for (int i = 0; v; v >>= 1, i++)
-    if (v & 1) s ^= m[i];
-  return s;
 }
 
 // One wave per 4 KiB page; 4 waves (256 threads) per workgroup.
@@ -240,13 +257,13 @@ __global__ void __launch_bounds__(256) k_crc32_pages(
     const uint32_t* __restrict__ buf, uint64_t npages,
     uint32_t* __restrict__ out) {
   __shared__ uint32_t s_tab[8][256];
-  __shared__ uint32_t s_mat[6][32];
+  __shared__ uint32_t s_stab[6][4][256];
 
-  // stage tables into LDS (8.75 KB)
+  // stage tables into LDS (8 KB slice-by-8 + 24 KB shift operators)
   for (uint32_t i = threadIdx.x; i < 8 * 256; i += blockDim.x)
     (&s_tab[0][0])[i] = (&c_crc_tab[0][0])[i];
-  for (uint32_t i = threadIdx.x; i < 6 * 32; i += blockDim.x)
-    (&s_mat[0][0])[i] = (&c_shift_mat[0][0])[i];
+  for (uint32_t i = threadIdx.x; i < 6 * 4 * 256; i += blockDim.x)
+    (&s_stab[0][0][0])[i] = (&c_shift_tab[0][0][0])[i];
   __syncthreads();
 
   const uint32_t lane = threadIdx.x & (WAVE - 1);
@@ -278,15 +295,22 @@ __global__ void __launch_bounds__(256) k_crc32_pages(
     }
     crc ^= 0xFFFFFFFFu;
 
-    // shift by tail = (63 - lane) * 64 bytes: apply M(64B<<k) per bit
-    uint32_t tail = (WAVE - 1) - lane;
+    // Log-tree combine: level k merges adjacent spans of 64B<<k.
+    // combine(cL, cR) = shift(cL, span) ^ cR; shift is linear over
+    // XOR, so the tree equals the flat XOR-of-shifted-lane-CRCs
+    // identity validated against zlib in tests/test_pattern.py.  The
+    // shift matrix is UNIFORM per level (byte-sliced: 4 LDS lookups),
+    // and the tree doubles as the wave reduction — every lane ends
+    // holding the page CRC.
 #pragma unroll
-    for (int k = 0; k < 6; k++)
-      if ((tail >> k) & 1) crc = lds_mat_times(s_mat[k], crc);
-
-    // XOR-reduce across the wave; lane 0 owns the page CRC
-    for (unsigned off = WAVE / 2; off; off >>= 1)
-      crc ^= __shfl_xor(crc, off, WAVE);
+    for (int k = 0; k < 6; k++) {
+      uint32_t partner = __shfl_xor(crc, 1u << k, WAVE);
+      bool left = ((lane >> k) & 1u) == 0;
+      uint32_t cl = left ? crc : partner;
+      uint32_t cr = left ? partner : crc;
+      crc = s_stab[k][0][cl & 0xff] ^ s_stab[k][1][(cl >> 8) & 0xff] ^
+            s_stab[k][2][(cl >> 16) & 0xff] ^ s_stab[k][3][cl >> 24] ^ cr;
+    }
     if (lane == 0) out[page] = crc;
   }
 }
@@ -317,7 +341,7 @@ extern "C" hipError_t rocp2p_verify(const void* buf, uint64_t nbytes,
                                     hipStream_t stream) {
   if (nbytes % 8) return hipErrorInvalidValue;
   uint64_t nwords = nbytes / 8;
-  uint32_t grid = stream_grid(nwords / 2, 256);
+  uint32_t grid = stream_grid(nwords / 8, 256);
   hipLaunchKernelGGL(k_verify, dim3(grid), dim3(256), 0, stream,
                      (const uint64_t*)buf, nwords, seed, d_mismatch);
   return hipGetLastError();
@@ -384,11 +408,11 @@ extern "C" hipError_t rocp2p_scatter(const void* src_base,
 extern "C" hipError_t rocp2p_crc32_init() {
   if (g_crc_ready) return hipSuccess;
   static uint32_t tab[8][256];
-  static uint32_t mats[6][32];
-  host_make_tables(tab, mats);
+  static uint32_t stab[6][4][256];
+  host_make_tables(tab, stab);
   hipError_t e = hipMemcpyToSymbol(HIP_SYMBOL(c_crc_tab), tab, sizeof(tab));
   if (e != hipSuccess) return e;
-  e = hipMemcpyToSymbol(HIP_SYMBOL(c_shift_mat), mats, sizeof(mats));
+  e = hipMemcpyToSymbol(HIP_SYMBOL(c_shift_tab), stab, sizeof(stab));
   if (e != hipSuccess) return e;
   g_crc_ready = true;
   return hipSuccess;
