@@ -156,3 +156,20 @@ def test_dmabuf_export(dev):
     st = os.fstat(fd)
     assert st is not None
     os.close(fd)
+
+
+def test_soak_gpu_short(dev):
+    from rocnrdma_amd.harness.soak import run_soak
+
+    stats = run_soak("sdma", secs=5.0, region_bytes=64 << 20, seed=3,
+                     device=dev)
+    assert stats["cycles"] > 0
+    assert stats["failures"] == 0
+
+
+def test_p2p_matrix_runs(dev):
+    from rocnrdma_amd.harness.xgmi_check import p2p_matrix
+
+    rows = p2p_matrix(mb=64, iters=3)
+    assert len(rows) >= 1  # 1 GPU -> self copy row
+    assert all(r["gbps"] > 50 for r in rows)

@@ -63,3 +63,12 @@ def test_python_lat_sweep():
     r = run_lat_point(tp, iters=200)
     assert r["mode"] == "lat"
     assert 0 < r["us_min"] <= r["us_p50"] <= r["us_p99"] <= r["us_max"]
+
+
+def test_soak_fake_short():
+    from rocnrdma_amd.harness.soak import run_soak
+
+    stats = run_soak("fake", secs=1.5, region_bytes=4 << 20, seed=7)
+    assert stats["cycles"] > 0
+    assert stats["audits"] == stats["cycles"]
+    assert stats["failures"] == 0
