@@ -1,0 +1,302 @@
+// SPDX-License-Identifier: MIT
+// In-process fake libibverbs (see infiniband/verbs.h in this dir).
+// Strict where it matters: the QP state machine requires the exact
+// attr-mask sets an mlx5 RC QP requires, MR access flags are enforced
+// on remote ops, and data movement happens at post_send time with
+// bounds checks — so the real verbs backend's connect/post/poll logic
+// is genuinely validated without an HCA.
+//
+// Env knobs:
+//   FAKE_VERBS_LINK=ib|eth   link layer reported by query_port
+//                            (default ib) — covers both AH paths.
+#include "infiniband/verbs.h"
+
+#include <cstdlib>
+#include <cstring>
+#include <cstdio>
+#include <deque>
+#include <vector>
+
+namespace {
+
+struct FakeCtx;
+
+struct FakeMr {
+  ibv_mr mr;
+  int access;
+  bool live;
+};
+
+struct FakeCq {
+  std::deque<ibv_wc> completions;
+};
+
+struct FakeQp {
+  ibv_qp qp;
+  FakeCq* send_cq;
+  int access_flags = 0;
+  uint32_t dest_qp_num = 0;
+  bool has_av = false;
+};
+
+struct FakeCtx {
+  std::vector<FakeMr*> mrs;
+  uint32_t next_key = 0x1000;
+  uint32_t next_qpn = 0x40;
+};
+
+FakeCtx g_ctx;
+char g_device_storage;
+ibv_device* g_device_list[2] = {
+    reinterpret_cast<ibv_device*>(&g_device_storage), nullptr};
+
+#define FAIL(msg)                                          \
+  do {                                                     \
+    fprintf(stderr, "fake_verbs: %s\n", msg);              \
+    return -1;                                             \
+  } while (0)
+
+FakeMr* find_mr_by_key(uint32_t key, bool remote) {
+  for (auto* m : g_ctx.mrs)
+    if (m->live && (remote ? m->mr.rkey : m->mr.lkey) == key) return m;
+  return nullptr;
+}
+
+bool range_ok(const FakeMr* m, uint64_t addr, uint64_t len) {
+  uint64_t base = (uint64_t)m->mr.addr;
+  return addr >= base && addr + len <= base + m->mr.length;
+}
+
+}  // namespace
+
+extern "C" {
+
+struct ibv_device** ibv_get_device_list(int* num) {
+  if (num) *num = 1;
+  return g_device_list;
+}
+
+void ibv_free_device_list(struct ibv_device**) {}
+
+struct ibv_context* ibv_open_device(struct ibv_device* d) {
+  return d == g_device_list[0] ? reinterpret_cast<ibv_context*>(&g_ctx)
+                               : nullptr;
+}
+
+int ibv_close_device(struct ibv_context*) { return 0; }
+
+struct ibv_pd* ibv_alloc_pd(struct ibv_context* c) {
+  return reinterpret_cast<ibv_pd*>(c);
+}
+
+int ibv_dealloc_pd(struct ibv_pd*) { return 0; }
+
+struct ibv_mr* ibv_reg_mr(struct ibv_pd* pd, void* addr, size_t length,
+                          int access) {
+  if (!pd || !addr || !length) return nullptr;
+  auto* m = new FakeMr();
+  m->mr.pd = pd;
+  m->mr.addr = addr;
+  m->mr.length = length;
+  m->mr.lkey = g_ctx.next_key++;
+  m->mr.rkey = g_ctx.next_key++;
+  m->access = access;
+  m->live = true;
+  g_ctx.mrs.push_back(m);
+  return &m->mr;
+}
+
+struct ibv_mr* ibv_reg_dmabuf_mr(struct ibv_pd* pd, uint64_t offset,
+                                 size_t length, uint64_t iova, int fd,
+                                 int access) {
+  // data plane for dmabuf regions is device memory — the fake records
+  // the registration (iova-addressed) but cannot memcpy it; posts
+  // against it fail loudly so tests keep to host MRs.
+  if (fd < 0) return nullptr;
+  (void)offset;
+  auto* m = new FakeMr();
+  m->mr.pd = pd;
+  m->mr.addr = (void*)iova;
+  m->mr.length = length;
+  m->mr.lkey = g_ctx.next_key++;
+  m->mr.rkey = g_ctx.next_key++;
+  m->access = access | (1 << 30);  // mark: not host-addressable
+  m->live = true;
+  g_ctx.mrs.push_back(m);
+  return &m->mr;
+}
+
+int ibv_dereg_mr(struct ibv_mr* mr) {
+  for (auto* m : g_ctx.mrs)
+    if (&m->mr == mr && m->live) {
+      m->live = false;
+      return 0;
+    }
+  return -1;
+}
+
+struct ibv_cq* ibv_create_cq(struct ibv_context*, int cqe, void*, void*,
+                             int) {
+  if (cqe <= 0) return nullptr;
+  return reinterpret_cast<ibv_cq*>(new FakeCq());
+}
+
+int ibv_destroy_cq(struct ibv_cq* cq) {
+  delete reinterpret_cast<FakeCq*>(cq);
+  return 0;
+}
+
+struct ibv_qp* ibv_create_qp(struct ibv_pd*, struct ibv_qp_init_attr* a) {
+  if (!a || !a->send_cq || a->qp_type != IBV_QPT_RC) return nullptr;
+  if (!a->cap.max_send_wr || !a->cap.max_send_sge) return nullptr;
+  auto* q = new FakeQp();
+  q->qp.qp_num = g_ctx.next_qpn++;
+  q->qp.state = IBV_QPS_RESET;
+  q->qp.qp_type = a->qp_type;
+  q->qp.send_cq = a->send_cq;
+  q->send_cq = reinterpret_cast<FakeCq*>(a->send_cq);
+  return &q->qp;
+}
+
+int ibv_destroy_qp(struct ibv_qp* qp) {
+  delete reinterpret_cast<FakeQp*>(qp);
+  return 0;
+}
+
+int ibv_query_port(struct ibv_context*, uint8_t port,
+                   struct ibv_port_attr* pa) {
+  if (port != 1 || !pa) return -1;
+  memset(pa, 0, sizeof(*pa));
+  pa->active_mtu = IBV_MTU_4096;
+  pa->max_mtu = IBV_MTU_4096;
+  const char* link = getenv("FAKE_VERBS_LINK");
+  if (link && !strcmp(link, "eth")) {
+    pa->link_layer = IBV_LINK_LAYER_ETHERNET;
+    pa->lid = 0;
+  } else {
+    pa->link_layer = IBV_LINK_LAYER_INFINIBAND;
+    pa->lid = 7;
+  }
+  pa->gid_tbl_len = 4;
+  return 0;
+}
+
+int ibv_query_gid(struct ibv_context*, uint8_t port, int index,
+                  union ibv_gid* gid) {
+  if (port != 1 || index < 0 || index >= 4 || !gid) return -1;
+  memset(gid, 0, sizeof(*gid));
+  gid->raw[15] = (uint8_t)(index + 1);
+  return 0;
+}
+
+// RC QP state machine with mlx5-grade required-mask checking.
+int ibv_modify_qp(struct ibv_qp* qp, struct ibv_qp_attr* a, int mask) {
+  auto* q = reinterpret_cast<FakeQp*>(qp);
+  if (!(mask & IBV_QP_STATE)) FAIL("modify_qp without IBV_QP_STATE");
+  switch (a->qp_state) {
+    case IBV_QPS_INIT: {
+      if (q->qp.state != IBV_QPS_RESET && q->qp.state != IBV_QPS_INIT)
+        FAIL("INIT from wrong state");
+      const int need = IBV_QP_STATE | IBV_QP_PKEY_INDEX | IBV_QP_PORT |
+                       IBV_QP_ACCESS_FLAGS;
+      if ((mask & need) != need) FAIL("RESET->INIT mask incomplete");
+      if (a->port_num != 1) FAIL("bad port");
+      q->access_flags = (int)a->qp_access_flags;
+      q->qp.state = IBV_QPS_INIT;
+      return 0;
+    }
+    case IBV_QPS_RTR: {
+      if (q->qp.state != IBV_QPS_INIT) FAIL("RTR from wrong state");
+      const int need = IBV_QP_STATE | IBV_QP_AV | IBV_QP_PATH_MTU |
+                       IBV_QP_DEST_QPN | IBV_QP_RQ_PSN |
+                       IBV_QP_MAX_DEST_RD_ATOMIC | IBV_QP_MIN_RNR_TIMER;
+      if ((mask & need) != need) FAIL("INIT->RTR mask incomplete");
+      if (a->path_mtu < IBV_MTU_256 || a->path_mtu > IBV_MTU_4096)
+        FAIL("bad mtu");
+      ibv_port_attr pa;
+      ibv_query_port(nullptr, 1, &pa);
+      if (pa.link_layer == IBV_LINK_LAYER_ETHERNET) {
+        if (!a->ah_attr.is_global) FAIL("RoCE needs GRH (is_global)");
+      } else {
+        if (!a->ah_attr.dlid) FAIL("IB needs dlid");
+      }
+      q->dest_qp_num = a->dest_qp_num;
+      q->has_av = true;
+      q->qp.state = IBV_QPS_RTR;
+      return 0;
+    }
+    case IBV_QPS_RTS: {
+      if (q->qp.state != IBV_QPS_RTR) FAIL("RTS from wrong state");
+      const int need = IBV_QP_STATE | IBV_QP_SQ_PSN | IBV_QP_TIMEOUT |
+                       IBV_QP_RETRY_CNT | IBV_QP_RNR_RETRY |
+                       IBV_QP_MAX_QP_RD_ATOMIC;
+      if ((mask & need) != need) FAIL("RTR->RTS mask incomplete");
+      q->qp.state = IBV_QPS_RTS;
+      return 0;
+    }
+    default:
+      FAIL("unsupported target state");
+  }
+}
+
+int ibv_post_send(struct ibv_qp* qp, struct ibv_send_wr* wr,
+                  struct ibv_send_wr** bad) {
+  auto* q = reinterpret_cast<FakeQp*>(qp);
+  for (; wr; wr = wr->next) {
+    if (bad) *bad = wr;
+    if (q->qp.state != IBV_QPS_RTS) FAIL("post_send: QP not in RTS");
+    if (wr->num_sge != 1) FAIL("post_send: expected 1 sge");
+    FakeMr* local = find_mr_by_key(wr->sg_list[0].lkey, false);
+    if (!local) FAIL("post_send: bad lkey");
+    if (!range_ok(local, wr->sg_list[0].addr, wr->sg_list[0].length))
+      FAIL("post_send: local sge out of MR bounds");
+    FakeMr* remote = find_mr_by_key(wr->wr.rdma.rkey, true);
+    if (!remote) FAIL("post_send: bad rkey");
+    if (!range_ok(remote, wr->wr.rdma.remote_addr, wr->sg_list[0].length))
+      FAIL("post_send: remote range out of MR bounds");
+    if (remote->access & (1 << 30))
+      FAIL("post_send: fake layer cannot address dmabuf (device) MRs");
+    if (wr->opcode == IBV_WR_RDMA_WRITE) {
+      if (!(remote->access & IBV_ACCESS_REMOTE_WRITE))
+        FAIL("post_send: remote MR lacks REMOTE_WRITE");
+      memcpy((void*)wr->wr.rdma.remote_addr, (void*)wr->sg_list[0].addr,
+             wr->sg_list[0].length);
+    } else if (wr->opcode == IBV_WR_RDMA_READ) {
+      if (!(remote->access & IBV_ACCESS_REMOTE_READ))
+        FAIL("post_send: remote MR lacks REMOTE_READ");
+      if (!(local->access & IBV_ACCESS_LOCAL_WRITE))
+        FAIL("post_send: local MR lacks LOCAL_WRITE");
+      memcpy((void*)wr->sg_list[0].addr, (void*)wr->wr.rdma.remote_addr,
+             wr->sg_list[0].length);
+    } else {
+      FAIL("post_send: unsupported opcode");
+    }
+    if (wr->send_flags & IBV_SEND_SIGNALED) {
+      ibv_wc wc;
+      memset(&wc, 0, sizeof(wc));
+      wc.wr_id = wr->wr_id;
+      wc.status = IBV_WC_SUCCESS;
+      wc.byte_len = wr->sg_list[0].length;
+      wc.qp_num = q->qp.qp_num;
+      q->send_cq->completions.push_back(wc);
+    }
+  }
+  if (bad) *bad = nullptr;
+  return 0;
+}
+
+int ibv_poll_cq(struct ibv_cq* cq, int n, struct ibv_wc* wc) {
+  auto* c = reinterpret_cast<FakeCq*>(cq);
+  int got = 0;
+  while (got < n && !c->completions.empty()) {
+    wc[got++] = c->completions.front();
+    c->completions.pop_front();
+  }
+  return got;
+}
+
+const char* ibv_wc_status_str(enum ibv_wc_status s) {
+  return s == IBV_WC_SUCCESS ? "success" : "error";
+}
+
+}  // extern "C"

@@ -79,3 +79,42 @@ def test_lat_mode(built):
     assert r["mode"] == "lat"
     assert 0 < r["us_min"] <= r["us_p50"] <= r["us_p99"] <= r["us_max"]
     assert r["integrity"] == "ok"
+
+
+FAKEVERBS = os.path.join(HARNESS, "build", "rocp2p_bw_fakeverbs")
+
+
+@pytest.fixture(scope="module")
+def built_fakeverbs(built):
+    assert os.path.exists(FAKEVERBS)
+    return FAKEVERBS
+
+
+@pytest.mark.parametrize("link", ["ib", "eth"])
+@pytest.mark.parametrize("direction", ["write", "read"])
+@pytest.mark.timeout(300)
+def test_verbs_backend_against_fake_layer(built_fakeverbs, link, direction):
+    """The REAL verbs backend (QP bring-up masks, MR access flags, WR
+    posting, CQ drain) validated by the strict in-process fake, on both
+    link-layer connect paths."""
+    env = dict(os.environ, FAKE_VERBS_LINK=link)
+    out = subprocess.run(
+        [built_fakeverbs, "--transport", "verbs", "--mr", "host", "--msg",
+         "65536", "--region", "2097152", "--secs", "0.1", "--dir",
+         direction, "--json"],
+        capture_output=True, text=True, timeout=120, env=env)
+    assert out.returncode == 0, out.stderr
+    r = json.loads(out.stdout.strip())
+    assert r["transport"] == "verbs"
+    assert r["integrity"] == "ok"
+
+
+@pytest.mark.timeout(300)
+def test_verbs_backend_lat_against_fake_layer(built_fakeverbs):
+    out = subprocess.run(
+        [built_fakeverbs, "--transport", "verbs", "--mr", "host", "--msg",
+         "4096", "--region", "1048576", "--lat", "300", "--json"],
+        capture_output=True, text=True, timeout=120)
+    assert out.returncode == 0, out.stderr
+    r = json.loads(out.stdout.strip())
+    assert r["mode"] == "lat" and r["integrity"] == "ok"
