@@ -50,6 +50,16 @@
 #include "rocnr_amd_rdma.h"
 #include "rocp2p_probe_abi.h"
 
+/* vm_flags_set() appeared in 6.3; earlier kernels mutate vm_flags
+ * directly (the shim provides its own definition). */
+#if !defined(__ROCNR_SHIM__) && LINUX_VERSION_CODE < KERNEL_VERSION(6, 3, 0)
+static inline void vm_flags_set(struct vm_area_struct *vma,
+				vm_flags_t flags)
+{
+	vma->vm_flags |= flags;
+}
+#endif
+
 MODULE_AUTHOR("ROCnRDMA-AMD project");
 MODULE_LICENSE("Dual MIT/GPL");
 MODULE_DESCRIPTION("raw amd_rdma probe device for MI355X GPU-direct RDMA");
