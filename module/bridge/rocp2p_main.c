@@ -387,10 +387,11 @@ static int rocp2p_dma_map(struct sg_table *sg_head, void *client_context,
 	 * mapping and is released in dma_unmap. */
 	*sg_head = map->sgt;
 	*nmap = (int)nsegs;
-	mutex_unlock(&reg->lock);
-
+	/* log before unlock: reg must not be touched once the lock drops
+	 * (a KFD revoke during registration can reach release()) */
 	rp_dbg("dma_map: va 0x%llx size 0x%llx -> %zu segs (kfd page %lu, max_seg 0x%llx)\n",
 	       reg->va, reg->size, nsegs, reg->page_size, max_seg);
+	mutex_unlock(&reg->lock);
 	return 0;
 }
 
