@@ -25,6 +25,7 @@
 
 #include <hip/hip_runtime.h>
 #include <cstdint>
+#include <cstdlib>
 #include "p2p_kernels.h"
 #include "p2p_pattern.h"
 
@@ -367,6 +368,19 @@ extern "C" hipError_t rocp2p_copy_nt(void* dst, const void* src,
   return hipGetLastError();
 }
 
+// The gather/scatter engine is PCIe-bound, not CU-bound; its grid is
+// capped separately (ROCP2P_GATHER_GRID env overrides) so concurrent
+// engines (full-duplex) can co-schedule instead of serializing on CUs.
+static uint32_t gather_grid_cap() {
+  static int cap = -1;
+  if (cap < 0) {
+    const char* e = getenv("ROCP2P_GATHER_GRID");
+    cap = e ? atoi(e) : 0;
+    if (cap <= 0) cap = 16384;
+  }
+  return (uint32_t)cap;
+}
+
 extern "C" hipError_t rocp2p_gather(void* dst_base,
                                     const uint64_t* d_dst_offs,
                                     const uint64_t* d_src_addrs,
@@ -377,6 +391,7 @@ extern "C" hipError_t rocp2p_gather(void* dst_base,
   uint64_t vecs_per_msg = msg_bytes / 16;
   uint64_t total = vecs_per_msg * n;
   uint32_t grid = stream_grid(total, 256);
+  if (grid > gather_grid_cap()) grid = gather_grid_cap();
   uint32_t vshift = (vecs_per_msg & (vecs_per_msg - 1))
                         ? 0xffffffffu
                         : (uint32_t)__builtin_ctzll(vecs_per_msg);
@@ -396,6 +411,7 @@ extern "C" hipError_t rocp2p_scatter(const void* src_base,
   uint64_t vecs_per_msg = msg_bytes / 16;
   uint64_t total = vecs_per_msg * n;
   uint32_t grid = stream_grid(total, 256);
+  if (grid > gather_grid_cap()) grid = gather_grid_cap();
   uint32_t vshift = (vecs_per_msg & (vecs_per_msg - 1))
                         ? 0xffffffffu
                         : (uint32_t)__builtin_ctzll(vecs_per_msg);
