@@ -67,3 +67,17 @@ def test_probe_cli_compiles(tmp_path):
          "-o", str(tmp_path / "cli")],
         capture_output=True, text=True)
     assert out.returncode == 0, out.stderr
+
+
+@pytest.mark.skipif(shutil.which("gcc") is None or
+                    shutil.which("make") is None, reason="no toolchain")
+@pytest.mark.timeout(900)
+@pytest.mark.parametrize("binary", ["bridge_tests_tsan", "probe_tests_tsan"])
+def test_shim_suites_tsan(binary):
+    """ThreadSanitizer builds of both race suites must be clean."""
+    subprocess.run(["make", "-C", SHIM, "all"], check=True,
+                   capture_output=True, text=True)
+    out = subprocess.run([os.path.join(SHIM, "build", binary)],
+                         capture_output=True, text=True, timeout=600)
+    assert out.returncode == 0, out.stdout + out.stderr
+    assert "WARNING: ThreadSanitizer" not in out.stdout + out.stderr
