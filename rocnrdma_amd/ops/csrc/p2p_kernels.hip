@@ -331,6 +331,7 @@ extern "C" hipError_t rocp2p_fill(void* buf, uint64_t nbytes, uint64_t seed,
   if (nbytes % 8) return hipErrorInvalidValue;
   uint64_t nwords = nbytes / 8;
   uint32_t grid = stream_grid(nwords / 16, 256);
+  if (grid > 8192) grid = 8192;  // fill peaks at 8192 WGs (hbm_bench)
   hipLaunchKernelGGL(k_fill, dim3(grid), dim3(256), 0, stream,
                      (uint64_t*)buf, nwords, seed);
   return hipGetLastError();
