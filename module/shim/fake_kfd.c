@@ -32,6 +32,7 @@ static struct fk_pin *fk_pins;
 static uint64_t fk_next_va = 0x700000000000ULL;
 static uint64_t fk_next_bus = 0xd000000000ULL;
 static long fk_stat_get, fk_stat_put, fk_stat_bad_put, fk_stat_cb;
+static int fk_fail_page_size;
 
 uint64_t fake_kfd_alloc(uint64_t size, unsigned int frag_every)
 {
@@ -240,12 +241,24 @@ static int fk_get_page_size(uint64_t address, uint64_t length,
 
 	(void)pid;
 	pthread_mutex_lock(&fk_lock);
+	if (fk_fail_page_size > 0) {
+		fk_fail_page_size--;
+		pthread_mutex_unlock(&fk_lock);
+		return -EIO;
+	}
 	ok = fk_find_locked(address) != NULL && length > 0;
 	pthread_mutex_unlock(&fk_lock);
 	if (!ok)
 		return -EINVAL;
 	*page_size = (unsigned long)FAKE_KFD_VRAM_PAGE;
 	return 0;
+}
+
+void fake_kfd_fail_page_size(int n)
+{
+	pthread_mutex_lock(&fk_lock);
+	fk_fail_page_size = n;
+	pthread_mutex_unlock(&fk_lock);
 }
 
 void fake_kfd_free(uint64_t va)

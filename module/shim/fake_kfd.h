@@ -46,6 +46,10 @@ long fake_kfd_put_pages_calls(void);
 long fake_kfd_bad_put_calls(void);	/* put of unknown pin = bridge bug */
 long fake_kfd_callbacks_fired(void);
 
+/* Fault injection: make the next N get_page_size calls fail (tests the
+ * bridge's documented 2 MiB fallback + warning path). */
+void fake_kfd_fail_page_size(int n);
+
 #ifdef __cplusplus
 }
 #endif

@@ -339,6 +339,26 @@ static void test_invalidate_race(void)
 	       ITERS);
 }
 
+static void test_page_size_fallback(void)
+{
+	/* KFD page-size query failing mid-registration must not fail the
+	 * MR: the bridge warns and reports the 2 MiB VRAM granule
+	 * (bridge policy, mirroring the reference's 4096 fallback at
+	 * amdp2p.c:334-340 but with the MI355X granule). */
+	uint64_t va = fake_kfd_alloc(8 * MiB, 0);
+	struct fake_ib_mr *mr = NULL;
+
+	fake_kfd_fail_page_size(2);	/* get_pages + get_page_size */
+	CHECK(fake_ib_reg_mr(va, 8 * MiB, &dev1, &mr) == 0);
+	CHECK(mr->page_size == 2 * MiB);
+	CHECK(fake_ib_dereg_mr(mr) == 0);
+	free(mr);
+	fake_kfd_free(va);
+	fake_kfd_fail_page_size(0);
+	check_balances(devs, 2);
+	printf("ok: page-size query failure falls back to 2 MiB\n");
+}
+
 static void test_huge_pin(void)
 {
 	/* 64 GiB pin (288 GB HBM sizing): 32768 chunks, hole every 1024 */
@@ -372,6 +392,7 @@ int main(void)
 	test_release_cleans_leftovers();
 	test_invalidate();
 	test_invalidate_race();
+	test_page_size_fallback();
 	test_huge_pin();
 
 	rocnr_shim_module_exit();
