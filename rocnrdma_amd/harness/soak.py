@@ -17,9 +17,13 @@ SIZES = [4 << 10, 16 << 10, 64 << 10, 256 << 10, 1 << 20, 4 << 20]
 
 def run_soak(transport: str = "auto", secs: float = 10.0,
              region_bytes: int = 256 << 20, seed: int = 1234,
-             device=None) -> dict:
+             device=None, metrics=None) -> dict:
     from rocnrdma_amd.transport import get_transport
 
+    if metrics is None:
+        from rocnrdma_amd.utils.metrics import TransferMetrics
+
+        metrics = TransferMetrics(port=None)
     rng = random.Random(seed)
     t_end = time.monotonic() + secs
     stats = {"cycles": 0, "msgs": 0, "bytes": 0, "audits": 0,
@@ -43,9 +47,13 @@ def run_soak(transport: str = "auto", secs: float = 10.0,
             stats["audits"] += 1
             if bad:
                 stats["failures"] += 1
+            moved = (posted + tp.msgs_per_region) * msg
             stats["msgs"] += posted + tp.msgs_per_region
-            stats["bytes"] += (posted + tp.msgs_per_region) * msg
+            stats["bytes"] += moved
             stats["cycles"] += 1
+            metrics.observe_bytes(direction, moved,
+                                  posted + tp.msgs_per_region)
+            metrics.observe_audit(ok=bad == 0)
         finally:
             tp.close()
     return stats
@@ -57,9 +65,14 @@ def main():
     ap.add_argument("--transport", default="auto")
     ap.add_argument("--region-bytes", type=int, default=256 << 20)
     ap.add_argument("--seed", type=int, default=1234)
+    ap.add_argument("--metrics-port", type=int, default=0,
+                    help="expose Prometheus metrics on this port")
     args = ap.parse_args()
+    from rocnrdma_amd.utils.metrics import TransferMetrics
+
+    metrics = TransferMetrics(port=args.metrics_port or None)
     stats = run_soak(args.transport, args.secs, args.region_bytes,
-                     args.seed)
+                     args.seed, metrics=metrics)
     print(stats)
     raise SystemExit(1 if stats["failures"] else 0)
 

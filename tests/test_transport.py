@@ -72,3 +72,16 @@ def test_soak_fake_short():
     assert stats["cycles"] > 0
     assert stats["audits"] == stats["cycles"]
     assert stats["failures"] == 0
+
+
+def test_metrics_capture_in_soak():
+    from rocnrdma_amd.harness.soak import run_soak
+    from rocnrdma_amd.utils.metrics import TransferMetrics
+
+    m = TransferMetrics(port=None)
+    stats = run_soak("fake", secs=1.0, region_bytes=4 << 20, seed=11,
+                     metrics=m)
+    text = m.render().decode()
+    assert "rocp2p_bytes_total" in text
+    assert "rocp2p_integrity_audits_total" in text
+    assert stats["failures"] == 0
