@@ -50,6 +50,7 @@
 #include <linux/scatterlist.h>
 #include <linux/dma-mapping.h>
 #include <linux/moduleparam.h>
+#include <linux/pci.h>
 
 #include "rocnr_peer_mem.h"
 #include "rocnr_amd_rdma.h"
@@ -88,9 +89,43 @@ static int rocp2p_param_get_a64(char *buf, const struct kernel_param *kp)
 static const struct kernel_param_ops rocp2p_a64_ops = {
 	.get = rocp2p_param_get_a64,
 };
+static atomic64_t rocp2p_bar_bytes = ATOMIC64_INIT(0);
 module_param_cb(active_regs, &rocp2p_a64_ops, &rocp2p_active_regs, 0444);
 module_param_cb(pinned_bytes, &rocp2p_a64_ops, &rocp2p_pinned_bytes, 0444);
 module_param_cb(invalidations, &rocp2p_a64_ops, &rocp2p_invalidations, 0444);
+module_param_cb(bar_bytes, &rocp2p_a64_ops, &rocp2p_bar_bytes, 0444);
+
+/* MI355X exposes all 288 GB of HBM3E through BAR0 when resizable/large
+ * BAR is enabled; PeerDirect DMA can only target VRAM inside the BAR
+ * window, so an undersized aperture silently truncates the peer-
+ * reachable region.  Validate at load (north star: "P2P BAR aperture
+ * ... chosen for 288 GB HBM per GPU"; RUNBOOK 'Restrictions').  The
+ * largest AMD display-class BAR0 is recorded in the bar_bytes module
+ * param for the liveness tooling. */
+static void rocp2p_check_bar_aperture(void)
+{
+	struct pci_dev *pdev = NULL;
+	u64 best = 0;
+
+	while ((pdev = pci_get_device(PCI_VENDOR_ID_ATI, PCI_ANY_ID,
+				      pdev)) != NULL) {
+		if ((pdev->class >> 16) == PCI_BASE_CLASS_DISPLAY) {
+			u64 len = pci_resource_len(pdev, 0);
+
+			if (len > best)
+				best = len;
+		}
+	}
+	atomic64_add((long long)best, &rocp2p_bar_bytes);
+	if (!best)
+		rp_warn("no AMD display-class PCI device visible\n");
+	else if (best < (16ULL << 30))
+		rp_warn("largest GPU BAR0 is %llu MiB — full-VRAM (large) BAR appears DISABLED; peer DMA window is truncated (see docs/RUNBOOK.md restrictions)\n",
+			best >> 20);
+	else
+		rp_info("GPU BAR0 aperture %llu GiB (full-VRAM BAR ok)\n",
+			best >> 30);
+}
 
 enum rocp2p_state {
 	ROCP2P_ACQUIRED = 0,	/* context exists, nothing pinned */
@@ -545,6 +580,8 @@ static int __init rocp2p_init(void)
 		sizeof(rocp2p_client.name));
 	strscpy(rocp2p_client.version, ROCP2P_DRIVER_VERSION,
 		sizeof(rocp2p_client.version));
+
+	rocp2p_check_bar_aperture();
 
 	ib_reg_handle = ib_register_peer_memory_client(&rocp2p_client,
 						       &ib_invalidate_cb);

@@ -325,6 +325,26 @@ static inline u64 dma_get_max_seg_size(struct device *dev)
 	return dev->max_seg ? dev->max_seg : 0x100000000ULL;
 }
 
+/* ---- pci (BAR aperture probe) ---- */
+#define PCI_VENDOR_ID_ATI 0x1002
+#define PCI_ANY_ID (~0)
+#define PCI_BASE_CLASS_DISPLAY 0x03
+struct pci_dev {
+	unsigned int vendor;
+	unsigned int device;
+	unsigned int class;	/* class<<16 | subclass<<8 | progif */
+	u64 bar_len[2];
+};
+struct pci_dev *pci_get_device(unsigned int vendor, unsigned int device,
+			       struct pci_dev *from);
+static inline u64 pci_resource_len(struct pci_dev *d, int bar)
+{
+	return (bar >= 0 && bar < 2) ? d->bar_len[bar] : 0;
+}
+static inline void pci_dev_put(struct pci_dev *d) { (void)d; }
+/* test control: install a fake device table (NULL, 0 to clear) */
+void rocnr_shim_set_pci_devices(struct pci_dev *devs, int n);
+
 /* ---- char device / uaccess / mmap (for the probe module) ---- */
 struct inode {
 	int i_dummy;

@@ -139,6 +139,32 @@ void dma_unmap_resource(struct device *dev, dma_addr_t addr, size_t size,
 	atomic64_dec(&dev->live_maps);
 }
 
+/* ---- pci ---- */
+static struct pci_dev *shim_pci_devs;
+static int shim_pci_n;
+
+void rocnr_shim_set_pci_devices(struct pci_dev *devs, int n)
+{
+	shim_pci_devs = devs;
+	shim_pci_n = n;
+}
+
+struct pci_dev *pci_get_device(unsigned int vendor, unsigned int device,
+			       struct pci_dev *from)
+{
+	int i = 0;
+
+	if (from)
+		i = (int)(from - shim_pci_devs) + 1;
+	for (; i < shim_pci_n; i++) {
+		if (shim_pci_devs[i].vendor == vendor &&
+		    (device == PCI_ANY_ID ||
+		     shim_pci_devs[i].device == device))
+			return &shim_pci_devs[i];
+	}
+	return 0;
+}
+
 /* ---- misc device ---- */
 static struct miscdevice *shim_misc;
 

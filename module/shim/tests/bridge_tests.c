@@ -375,6 +375,29 @@ static void test_huge_pin(void)
 	printf("ok: 64 GiB pin -> 32 sg entries\n");
 }
 
+static void test_bar_aperture_probe(void)
+{
+	/* init already probed an empty PCI table (warned).  Re-run module
+	 * init against a small-BAR and a large-BAR topology and make sure
+	 * both load fine (the check warns, never fails the load). */
+	static struct pci_dev devs[] = {
+		{ .vendor = 0x1002, .device = 0x75a3,
+		  .class = 0x030000, .bar_len = { 256ULL << 20, 2 << 20 } },
+		{ .vendor = 0x8086, .device = 0x1234,
+		  .class = 0x020000, .bar_len = { 1 << 20, 0 } },
+	};
+
+	rocnr_shim_module_exit();
+	rocnr_shim_set_pci_devices(devs, 2);
+	CHECK(rocnr_shim_module_init() == 0);	/* small BAR: warns, loads */
+	rocnr_shim_module_exit();
+
+	devs[0].bar_len[0] = 288ULL << 30;	/* full-VRAM BAR */
+	CHECK(rocnr_shim_module_init() == 0);
+	rocnr_shim_set_pci_devices(0, 0);
+	printf("ok: BAR aperture probe (small warns, large ok, load never fails)\n");
+}
+
 int main(void)
 {
 	test_coalesce_unit();
@@ -394,6 +417,7 @@ int main(void)
 	test_invalidate_race();
 	test_page_size_fallback();
 	test_huge_pin();
+	test_bar_aperture_probe();
 
 	rocnr_shim_module_exit();
 	CHECK(fake_ib_client() == NULL);

@@ -10,7 +10,11 @@ fi
 a=$(cat "$P/active_regs")
 b=$(cat "$P/pinned_bytes")
 i=$(cat "$P/invalidations")
-echo "rocp2p: active_regs=$a pinned_bytes=$b invalidations=$i"
+bar=$(cat "$P/bar_bytes" 2>/dev/null || echo "?")
+echo "rocp2p: active_regs=$a pinned_bytes=$b invalidations=$i bar_bytes=$bar"
+if [ "$bar" != "?" ] && [ "$bar" -lt 17179869184 ] 2>/dev/null; then
+    echo "rocp2p: WARNING: GPU BAR0 aperture < 16 GiB — large BAR disabled?"
+fi
 if [ "$a" = "0" ] && [ "$b" = "0" ]; then
     echo "rocp2p: loaded but no registrations have dispatched here."
     echo "  -> run: harness/build/rocp2p_bw --transport verbs --mr peer"
