@@ -13,7 +13,7 @@ from .base import Transport
 
 
 def available_transports() -> list[str]:
-    names = ["fake"]
+    names = ["fake", "shm"]
     try:
         import torch
 
@@ -45,4 +45,8 @@ def get_transport(name: str, **kw) -> Transport:
         from .verbs import VerbsTransport
 
         return VerbsTransport(**kw)
+    if name == "shm":
+        from .shm import ShmInitiatorTransport
+
+        return ShmInitiatorTransport(**kw)
     raise ValueError(f"unknown transport {name!r}")
