@@ -58,4 +58,14 @@ std::unique_ptr<Transport> make_transport(const std::string& name,
 bool hip_available();
 bool verbs_runtime_available();  // libibverbs loads AND >=1 device
 
+// Client/server mode (ib_write_bw shape): the target registers the
+// region and answers OOB control ops; the client performs one-sided
+// ops against it.  Implemented in verbs_backend.cpp; throws on
+// verbs-less builds.
+std::unique_ptr<Transport> make_verbs_client(const TransportConfig& cfg,
+                                             const std::string& host,
+                                             int port);
+int run_verbs_target(const TransportConfig& cfg, int port,
+                     void (*announce)(int));
+
 }  // namespace rocp2p

@@ -11,7 +11,10 @@
 // --gpus N fans out one transport ("QP") per GPU on N worker threads and
 // reports per-GPU + aggregate bandwidth (BASELINE configs 4-5).
 #include <algorithm>
+#include <atomic>
 #include <chrono>
+#include <condition_variable>
+#include <mutex>
 #include <cstdio>
 #include <cstdlib>
 #include <cstring>
@@ -109,6 +112,9 @@ int main(int argc, char** argv) {
   int gpus = 1;
   bool sweep = false, json = false, integrity = true, bidir = false;
   int lat_iters = 0;
+  int serve_port = -1;
+  bool remote_selftest = false;
+  std::string connect_to;
   uint64_t seed = 0xC0FFEE;
 
   for (int i = 1; i < argc; i++) {
@@ -134,6 +140,9 @@ int main(int argc, char** argv) {
     else if (a == "--inflight") cfg.inflight = strtoull(next().c_str(), 0, 0);
     else if (a == "--seed") seed = strtoull(next().c_str(), 0, 0);
     else if (a == "--wc") cfg.wc_staging = true;
+    else if (a == "--serve") serve_port = atoi(next().c_str());
+    else if (a == "--connect") connect_to = next();
+    else if (a == "--remote-selftest") remote_selftest = true;
     else if (a == "--lat") lat_iters = atoi(next().c_str());
     else if (a == "--sweep") sweep = true;
     else if (a == "--json") json = true;
@@ -142,6 +151,62 @@ int main(int argc, char** argv) {
       fprintf(stderr, "unknown arg %s\n", a.c_str());
       return 2;
     }
+  }
+
+  if (serve_port >= 0) {
+    // passive target (ib_write_bw server shape); blocks until the
+    // client says bye
+    run_verbs_target(cfg, serve_port,
+                     [](int p) { printf("listening on port %d\n", p); });
+    return 0;
+  }
+
+  if (remote_selftest || !connect_to.empty()) {
+    // client/server data plane through the verbs backend.  selftest:
+    // both endpoints in THIS process over real TCP loopback (runs
+    // under the fake-verbs CI layer; on an HCA host it exercises the
+    // real NIC end to end on one box).
+    std::unique_ptr<Transport> tp;
+    std::thread server_thread;
+    if (remote_selftest) {
+      static std::atomic<int> s_port{0};
+      static std::mutex s_mu;
+      static std::condition_variable s_cv;
+      TransportConfig scfg = cfg;
+      server_thread = std::thread([scfg] {
+        run_verbs_target(scfg, 0, [](int p) {
+          {
+            std::lock_guard<std::mutex> g(s_mu);
+            s_port.store(p);
+          }
+          s_cv.notify_all();
+        });
+      });
+      {
+        std::unique_lock<std::mutex> lk(s_mu);
+        s_cv.wait(lk, [] { return s_port.load() != 0; });
+      }
+      tp = make_verbs_client(cfg, "127.0.0.1", s_port.load());
+    } else {
+      auto colon = connect_to.rfind(':');
+      if (colon == std::string::npos) {
+        fprintf(stderr, "--connect needs HOST:PORT\n");
+        return 2;
+      }
+      tp = make_verbs_client(cfg, connect_to.substr(0, colon),
+                             atoi(connect_to.c_str() + colon + 1));
+    }
+    Result r = lat_iters ? run_lat(*tp, lat_iters, seed, integrity)
+                         : run_point(*tp, secs, seed, integrity);
+    const char* ok = !integrity ? "skipped" : (r.bad == 0 ? "ok" : "FAILED");
+    printf("{\"transport\":\"%s\",\"msg_bytes\":%zu,\"mode\":\"%s\","
+           "\"gbps\":%.3f,\"msgs\":%llu,\"remote_integrity\":\"%s\"}\n",
+           tp->name(), cfg.msg_bytes,
+           remote_selftest ? "remote-selftest" : "client", r.gbps,
+           (unsigned long long)r.msgs, ok);
+    tp.reset();  // sends nothing; dtor closes OOB -> server exits loop
+    if (server_thread.joinable()) server_thread.join();
+    return (integrity && r.bad) ? 1 : 0;
   }
 
   std::vector<size_t> sizes =

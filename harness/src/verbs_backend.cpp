@@ -1,21 +1,28 @@
 // SPDX-License-Identifier: MIT
 // IB-verbs backend — the REAL PeerDirect data path: ibv_reg_mr on a
 // hipMalloc pointer (dispatched to the rocp2p bridge by the IB core's
-// peer-memory probe) and one-sided RDMA WRITE/READ through a loopback
-// RC QP pair, perftest-style.  MR modes:
-//   peer   — ibv_reg_mr(pd, hipMalloc ptr, ...): requires rocp2p.ko
-//            (the product under test);
-//   dmabuf — ibv_reg_dmabuf_mr over hipMemGetHandleForAddressRange:
-//            the modern kernel-module-free path, used as cross-check
-//            (SURVEY.md §5 "dmabuf is a validation path, not the design");
-//   host   — plain host memory (BASELINE config 1 loopback).
+// peer-memory probe) and one-sided RDMA WRITE/READ, perftest-style.
 //
-// This pool ships no rdma-core, so the implementation is compile-gated
-// on <infiniband/verbs.h>; on verbs-less hosts a stub throws with an
-// actionable message and verbs_runtime_available() (dlopen probe)
-// reports false.  The gated code follows the stable documented verbs
-// API; it has NOT run against real hardware from this pool — treat the
-// first run on an HCA host as a bring-up step (docs/RUNBOOK.md).
+// Modes:
+//  - loopback (default): an RC QP pair inside one process (BASELINE
+//    configs 1-3);
+//  - client/server: two endpoints exchange QP/MR parameters over the
+//    TCP OOB bootstrap (rocp2p_oob.h — the C++ twin of
+//    rocnrdma_amd/transport/oob.py) and the client performs one-sided
+//    ops against the server's registered region, with REMOTE
+//    verification over the control socket (ib_write_bw server/client
+//    shape; works across two hosts on HCA-equipped machines).
+//
+// MR modes: peer (ibv_reg_mr on hipMalloc VA — requires rocp2p.ko, the
+// product under test), dmabuf (hipMemGetHandleForAddressRange +
+// ibv_reg_dmabuf_mr — module-free cross-check), host.
+//
+// This pool ships no rdma-core, so everything is compile-gated on
+// <infiniband/verbs.h>; the fake-verbs CI layer (harness/fakeverbs/)
+// compiles and executes THIS file's loopback and client/server logic
+// in-process, including the OOB exchange and remote QP bring-up
+// (tests/test_native_harness.py).  First run on a real HCA is a
+// bring-up step (docs/RUNBOOK.md).
 #include <dlfcn.h>
 
 #include <cstring>
@@ -27,6 +34,9 @@
 namespace rocp2p {
 
 bool verbs_runtime_available() {
+#ifdef ROCNR_FAKE_VERBS
+  return true;  // the fake layer is linked in
+#else
   void* h = dlopen("libibverbs.so.1", RTLD_NOW | RTLD_LOCAL);
   if (!h) return false;
   using get_list_t = void** (*)(int*);
@@ -42,6 +52,7 @@ bool verbs_runtime_available() {
   }
   dlclose(h);
   return ok;
+#endif
 }
 
 }  // namespace rocp2p
@@ -56,107 +67,319 @@ bool verbs_runtime_available() {
 #include <infiniband/verbs.h>
 #include <hip/hip_runtime.h>
 
+#include <cstdio>
 #include <vector>
 
+#include "rocp2p_oob.h"
 #include "../../rocnrdma_amd/ops/csrc/p2p_kernels.h"
 #include "../../rocnrdma_amd/ops/csrc/p2p_pattern.h"
 
 namespace rocp2p {
 
-#define VB_THROW(cond, msg)                                          \
-  do {                                                               \
+#define VB_THROW(cond, msg)                                              \
+  do {                                                                   \
     if (!(cond)) throw std::runtime_error(std::string("verbs: ") + msg); \
   } while (0)
 
+namespace {
+
+// ---- small shared pieces -------------------------------------------
+
+struct PeerInfo {
+  uint32_t qpn = 0;
+  uint16_t lid = 0;
+  uint8_t gid[16] = {0};
+  int mtu = IBV_MTU_4096;
+};
+
+std::string gid_hex(const uint8_t* g) {
+  char buf[33];
+  for (int i = 0; i < 16; i++) sprintf(buf + 2 * i, "%02x", g[i]);
+  return std::string(buf, 32);
+}
+
+void gid_unhex(const std::string& s, uint8_t* g) {
+  VB_THROW(s.size() == 32, "bad gid encoding");
+  for (int i = 0; i < 16; i++)
+    g[i] = (uint8_t)strtoul(s.substr(2 * i, 2).c_str(), nullptr, 16);
+}
+
+ibv_context* open_first_device() {
+  int ndev = 0;
+  ibv_device** devs = ibv_get_device_list(&ndev);
+  VB_THROW(devs && ndev > 0, "no IB devices");
+  ibv_context* ctx = ibv_open_device(devs[0]);
+  ibv_free_device_list(devs);
+  VB_THROW(ctx, "ibv_open_device failed");
+  return ctx;
+}
+
+ibv_qp* make_rc_qp(ibv_pd* pd, ibv_cq* cq, size_t inflight) {
+  ibv_qp_init_attr a;
+  memset(&a, 0, sizeof(a));
+  a.send_cq = cq;
+  a.recv_cq = cq;
+  a.qp_type = IBV_QPT_RC;
+  a.cap.max_send_wr = (uint32_t)(2 * inflight + 8);
+  a.cap.max_recv_wr = 16;
+  a.cap.max_send_sge = 1;
+  a.cap.max_recv_sge = 1;
+  ibv_qp* qp = ibv_create_qp(pd, &a);
+  VB_THROW(qp, "ibv_create_qp failed");
+  return qp;
+}
+
+PeerInfo local_info(ibv_context* ctx, ibv_qp* qp, int gid_index) {
+  ibv_port_attr pattr;
+  VB_THROW(ibv_query_port(ctx, 1, &pattr) == 0, "ibv_query_port");
+  PeerInfo pi;
+  pi.qpn = qp->qp_num;
+  pi.lid = pattr.lid;
+  pi.mtu = (int)pattr.active_mtu;
+  if (pattr.link_layer == IBV_LINK_LAYER_ETHERNET) {
+    union ibv_gid gid;
+    VB_THROW(ibv_query_gid(ctx, 1, gid_index, &gid) == 0, "ibv_query_gid");
+    memcpy(pi.gid, gid.raw, 16);
+  }
+  return pi;
+}
+
+// Bring an RC QP to RTS against `peer` (mlx5-grade mask sets — the
+// fake-verbs layer validates them strictly).
+void qp_to_rts(ibv_context* ctx, ibv_qp* qp, const PeerInfo& peer,
+               int gid_index) {
+  ibv_port_attr pattr;
+  VB_THROW(ibv_query_port(ctx, 1, &pattr) == 0, "ibv_query_port");
+
+  ibv_qp_attr at;
+  memset(&at, 0, sizeof(at));
+  at.qp_state = IBV_QPS_INIT;
+  at.pkey_index = 0;
+  at.port_num = 1;
+  at.qp_access_flags = IBV_ACCESS_LOCAL_WRITE | IBV_ACCESS_REMOTE_WRITE |
+                       IBV_ACCESS_REMOTE_READ;
+  VB_THROW(ibv_modify_qp(qp, &at,
+                         IBV_QP_STATE | IBV_QP_PKEY_INDEX | IBV_QP_PORT |
+                             IBV_QP_ACCESS_FLAGS) == 0,
+           "modify->INIT");
+
+  memset(&at, 0, sizeof(at));
+  at.qp_state = IBV_QPS_RTR;
+  at.path_mtu = (enum ibv_mtu)peer.mtu;
+  at.dest_qp_num = peer.qpn;
+  at.rq_psn = 0;
+  at.max_dest_rd_atomic = 4;
+  at.min_rnr_timer = 12;
+  at.ah_attr.port_num = 1;
+  if (pattr.link_layer == IBV_LINK_LAYER_ETHERNET) {
+    at.ah_attr.is_global = 1;
+    memcpy(at.ah_attr.grh.dgid.raw, peer.gid, 16);
+    at.ah_attr.grh.sgid_index = (uint8_t)gid_index;
+    at.ah_attr.grh.hop_limit = 64;
+  } else {
+    at.ah_attr.is_global = 0;
+    at.ah_attr.dlid = peer.lid;
+  }
+  VB_THROW(ibv_modify_qp(qp, &at,
+                         IBV_QP_STATE | IBV_QP_AV | IBV_QP_PATH_MTU |
+                             IBV_QP_DEST_QPN | IBV_QP_RQ_PSN |
+                             IBV_QP_MAX_DEST_RD_ATOMIC |
+                             IBV_QP_MIN_RNR_TIMER) == 0,
+           "modify->RTR");
+
+  memset(&at, 0, sizeof(at));
+  at.qp_state = IBV_QPS_RTS;
+  at.sq_psn = 0;
+  at.timeout = 14;
+  at.retry_cnt = 7;
+  at.rnr_retry = 7;
+  at.max_rd_atomic = 4;
+  VB_THROW(ibv_modify_qp(qp, &at,
+                         IBV_QP_STATE | IBV_QP_SQ_PSN | IBV_QP_TIMEOUT |
+                             IBV_QP_RETRY_CNT | IBV_QP_RNR_RETRY |
+                             IBV_QP_MAX_QP_RD_ATOMIC) == 0,
+           "modify->RTS");
+}
+
+// A registered region: host malloc, GPU peer (ibv_reg_mr on hipMalloc
+// VA — the bridge path) or GPU dmabuf.
+struct Region {
+  uint8_t* host = nullptr;
+  uint8_t* gpu = nullptr;
+  uint8_t* ptr = nullptr;
+  ibv_mr* mr = nullptr;
+  size_t bytes = 0;
+  std::string mode;
+
+  void create(ibv_pd* pd, size_t nbytes, std::string mr_mode,
+              int device_index) {
+    bytes = nbytes;
+    mode = mr_mode;
+    if (mode == "auto") mode = hip_available() ? "peer" : "host";
+    int acc = IBV_ACCESS_LOCAL_WRITE | IBV_ACCESS_REMOTE_WRITE |
+              IBV_ACCESS_REMOTE_READ;
+    if (mode == "host") {
+      host = (uint8_t*)aligned_alloc(4096, nbytes);
+      VB_THROW(host, "region alloc failed");
+      memset(host, 0, nbytes);
+      ptr = host;
+      mr = ibv_reg_mr(pd, ptr, nbytes, acc);
+      VB_THROW(mr, "ibv_reg_mr(region) failed");
+      return;
+    }
+    VB_THROW(hipSetDevice(device_index) == hipSuccess, "hipSetDevice");
+    VB_THROW(hipMalloc((void**)&gpu, nbytes) == hipSuccess,
+             "hipMalloc region");
+    ptr = gpu;
+    if (mode == "dmabuf") {
+      int fd = -1;
+      VB_THROW(hipMemGetHandleForAddressRange(
+                   &fd, gpu, nbytes, hipMemRangeHandleTypeDmaBufFd, 0) ==
+                   hipSuccess,
+               "hipMemGetHandleForAddressRange(dmabuf) failed");
+      mr = ibv_reg_dmabuf_mr(pd, 0, nbytes, (uint64_t)gpu, fd, acc);
+      VB_THROW(mr, "ibv_reg_dmabuf_mr failed");
+    } else {
+      // peer mode: dispatched to the rocp2p bridge by the IB core
+      mr = ibv_reg_mr(pd, ptr, nbytes, acc);
+      VB_THROW(mr, "ibv_reg_mr(GPU VA) failed — rocp2p bridge loaded?");
+    }
+  }
+
+  uint64_t verify(uint64_t seed) {
+    if (gpu) {
+      unsigned long long* d_bad = nullptr;
+      VB_THROW(hipMalloc((void**)&d_bad, 8) == hipSuccess, "hipMalloc");
+      (void)hipMemset(d_bad, 0, 8);
+      VB_THROW(rocp2p_verify(gpu, bytes, seed, d_bad, 0) == hipSuccess,
+               "verify kernel");
+      unsigned long long h_bad = 0;
+      (void)hipMemcpy(&h_bad, d_bad, 8, hipMemcpyDeviceToHost);
+      (void)hipFree(d_bad);
+      return h_bad;
+    }
+    const uint64_t* r = reinterpret_cast<const uint64_t*>(host);
+    uint64_t bad = 0;
+    for (size_t w = 0; w < bytes / 8; w++)
+      bad += (r[w] != rocp2p_pattern_word(seed, w));
+    return bad;
+  }
+
+  void fill(uint64_t seed) {
+    if (gpu) {
+      VB_THROW(rocp2p_fill(gpu, bytes, seed, 0) == hipSuccess,
+               "fill kernel");
+      (void)hipDeviceSynchronize();
+      return;
+    }
+    uint64_t* r = reinterpret_cast<uint64_t*>(host);
+    for (size_t w = 0; w < bytes / 8; w++)
+      r[w] = rocp2p_pattern_word(seed, w);
+  }
+
+  void destroy() {
+    if (mr) ibv_dereg_mr(mr);
+    if (host) free(host);
+    if (gpu) (void)hipFree(gpu);
+    mr = nullptr;
+    host = gpu = ptr = nullptr;
+  }
+};
+
+int env_gid_index() {
+  const char* e = getenv("ROCP2P_GID_INDEX");
+  return e ? atoi(e) : 1;
+}
+
+}  // namespace
+
+// ---- loopback + client transports ----------------------------------
+
 class VerbsTransport final : public Transport {
  public:
-  explicit VerbsTransport(const TransportConfig& cfg) : Transport(cfg) {
+  // mode: loopback (oob == nullptr) or client (remote target via oob)
+  VerbsTransport(const TransportConfig& cfg, OobSocket* oob)
+      : Transport(cfg), oob_(oob) {
     if (cfg.region_bytes % cfg.msg_bytes)
       throw std::runtime_error("region must be a multiple of msg size");
     inflight_ = cfg.inflight ? cfg.inflight : 64;
     if (inflight_ > msgs_per_region()) inflight_ = msgs_per_region();
 
-    int ndev = 0;
-    ibv_device** devs = ibv_get_device_list(&ndev);
-    VB_THROW(devs && ndev > 0, "no IB devices");
-    ctx_ = ibv_open_device(devs[0]);
-    ibv_free_device_list(devs);
-    VB_THROW(ctx_, "ibv_open_device failed");
-
+    ctx_ = open_first_device();
     pd_ = ibv_alloc_pd(ctx_);
     VB_THROW(pd_, "ibv_alloc_pd failed");
     cq_ = ibv_create_cq(ctx_, 2 * (int)inflight_ + 16, nullptr, nullptr, 0);
     VB_THROW(cq_, "ibv_create_cq failed");
 
-    // staging: host-pinned (hip) so the same buffers work for GPU paths
-    mr_mode_ = cfg.verbs_mr;
-    if (mr_mode_ == "auto") mr_mode_ = hip_available() ? "peer" : "host";
-    if (mr_mode_ == "host") {
-      staging_ = (uint8_t*)aligned_alloc(4096, inflight_ * cfg.msg_bytes);
-      region_host_ = (uint8_t*)aligned_alloc(4096, cfg.region_bytes);
-      region_ptr_ = region_host_;
-    } else {
+    // staging: host-pinned (hip) so the same buffers serve GPU paths
+    bool use_hip = hip_available();
+    if (use_hip) {
       VB_THROW(hipSetDevice(cfg.device_index) == hipSuccess, "hipSetDevice");
       VB_THROW(hipHostMalloc((void**)&staging_,
                              inflight_ * cfg.msg_bytes, 0) == hipSuccess,
                "hipHostMalloc staging");
-      VB_THROW(hipMalloc((void**)&region_gpu_, cfg.region_bytes) ==
-                   hipSuccess,
-               "hipMalloc region");
-      region_ptr_ = region_gpu_;
+      staging_hip_ = true;
+    } else {
+      staging_ = (uint8_t*)aligned_alloc(4096, inflight_ * cfg.msg_bytes);
+      VB_THROW(staging_, "staging alloc failed");
     }
-
     staging_mr_ = ibv_reg_mr(pd_, staging_, inflight_ * cfg.msg_bytes,
                              IBV_ACCESS_LOCAL_WRITE);
     VB_THROW(staging_mr_, "ibv_reg_mr(staging) failed");
 
-    int acc = IBV_ACCESS_LOCAL_WRITE | IBV_ACCESS_REMOTE_WRITE |
-              IBV_ACCESS_REMOTE_READ;
-    if (mr_mode_ == "dmabuf") {
-#ifdef IBV_ACCESS_RELAXED_ORDERING
-      // optional; not required for correctness
-#endif
-      int fd = -1;
-      // export the HBM range as a dmabuf (ROCm >= 5.7)
-      VB_THROW(hipMemGetHandleForAddressRange(
-                   &fd, region_gpu_, cfg.region_bytes,
-                   hipMemRangeHandleTypeDmaBufFd, 0) == hipSuccess,
-               "hipMemGetHandleForAddressRange(dmabuf) failed");
-      region_mr_ = ibv_reg_dmabuf_mr(pd_, 0, cfg.region_bytes,
-                                     (uint64_t)region_gpu_, fd, acc);
-      VB_THROW(region_mr_, "ibv_reg_dmabuf_mr failed");
-    } else {
-      // peer mode: the IB core's peer-memory probe must dispatch this
-      // GPU VA to the rocp2p bridge; failure here on a GPU pointer
-      // means the bridge is not loaded/registered (RUNBOOK.md).
-      region_mr_ = ibv_reg_mr(pd_, region_ptr_, cfg.region_bytes, acc);
-      VB_THROW(region_mr_,
-               mr_mode_ == "peer"
-                   ? "ibv_reg_mr(GPU VA) failed — rocp2p bridge loaded?"
-                   : "ibv_reg_mr(region) failed");
+    qp_ = make_rc_qp(pd_, cq_, inflight_);
+
+    if (!oob_) {
+      // loopback: local region + a second QP as the passive end
+      region_.create(pd_, cfg.region_bytes, cfg.verbs_mr, cfg.device_index);
+      remote_addr_ = (uint64_t)region_.ptr;
+      remote_rkey_ = region_.mr->rkey;
+      qp_peer_ = make_rc_qp(pd_, cq_, inflight_);
+      PeerInfo a = local_info(ctx_, qp_, env_gid_index());
+      PeerInfo b = local_info(ctx_, qp_peer_, env_gid_index());
+      qp_to_rts(ctx_, qp_, b, env_gid_index());
+      qp_to_rts(ctx_, qp_peer_, a, env_gid_index());
+      return;
     }
 
-    qp_send_ = make_qp();
-    qp_recv_ = make_qp();
-    connect_loopback(qp_send_, qp_recv_);
-    connect_loopback(qp_recv_, qp_send_);
+    // client: exchange with the remote target
+    PeerInfo mine = local_info(ctx_, qp_, env_gid_index());
+    oob_->send_kv({{"qpn", std::to_string(mine.qpn)},
+                   {"lid", std::to_string(mine.lid)},
+                   {"gid", gid_hex(mine.gid)},
+                   {"mtu", std::to_string(mine.mtu)},
+                   {"region", std::to_string(cfg.region_bytes)}});
+    KvMap srv = oob_->recv_kv();
+    PeerInfo peer;
+    peer.qpn = (uint32_t)strtoul(srv.at("qpn").c_str(), nullptr, 10);
+    peer.lid = (uint16_t)strtoul(srv.at("lid").c_str(), nullptr, 10);
+    gid_unhex(srv.at("gid"), peer.gid);
+    peer.mtu = atoi(srv.at("mtu").c_str());
+    remote_addr_ = strtoull(srv.at("raddr").c_str(), nullptr, 10);
+    remote_rkey_ = (uint32_t)strtoul(srv.at("rkey").c_str(), nullptr, 10);
+    uint64_t rbytes = strtoull(srv.at("rbytes").c_str(), nullptr, 10);
+    VB_THROW(rbytes >= cfg.region_bytes, "target region too small");
+    qp_to_rts(ctx_, qp_, peer, env_gid_index());
   }
 
   ~VerbsTransport() override {
-    if (qp_send_) ibv_destroy_qp(qp_send_);
-    if (qp_recv_) ibv_destroy_qp(qp_recv_);
-    if (region_mr_) ibv_dereg_mr(region_mr_);
+    if (qp_) ibv_destroy_qp(qp_);
+    if (qp_peer_) ibv_destroy_qp(qp_peer_);
+    region_.destroy();
     if (staging_mr_) ibv_dereg_mr(staging_mr_);
     if (cq_) ibv_destroy_cq(cq_);
     if (pd_) ibv_dealloc_pd(pd_);
     if (ctx_) ibv_close_device(ctx_);
-    if (region_host_) free(region_host_);
-    if (region_gpu_) hipFree(region_gpu_);
-    if (mr_mode_ == "host") free(staging_);
-    else if (staging_) hipHostFree(staging_);
+    if (staging_) {
+      if (staging_hip_) (void)hipHostFree(staging_);
+      else free(staging_);
+    }
+    delete oob_;
   }
 
-  const char* name() const override { return "verbs"; }
+  const char* name() const override {
+    return oob_ ? "verbs-client" : "verbs";
+  }
 
   void post_many(uint64_t start, uint64_t n) override {
     for (uint64_t i = start; i < start + n; i++) {
@@ -174,10 +397,10 @@ class VerbsTransport final : public Transport {
                                                : IBV_WR_RDMA_READ;
       wr.send_flags = IBV_SEND_SIGNALED;
       wr.wr.rdma.remote_addr =
-          (uint64_t)region_ptr_ + (i % msgs_per_region()) * cfg_.msg_bytes;
-      wr.wr.rdma.rkey = region_mr_->rkey;
+          remote_addr_ + (i % msgs_per_region()) * cfg_.msg_bytes;
+      wr.wr.rdma.rkey = remote_rkey_;
       ibv_send_wr* bad = nullptr;
-      VB_THROW(ibv_post_send(qp_send_, &wr, &bad) == 0, "ibv_post_send");
+      VB_THROW(ibv_post_send(qp_, &wr, &bad) == 0, "ibv_post_send");
       outstanding_++;
     }
   }
@@ -186,7 +409,6 @@ class VerbsTransport final : public Transport {
 
   uint64_t integrity_check(uint64_t seed) override {
     const size_t words_per_msg = cfg_.msg_bytes / 8;
-    const bool gpu = region_gpu_ != nullptr;
     uint64_t bad = 0;
     if (cfg_.dir == Direction::Write) {
       for (size_t base = 0; base < msgs_per_region(); base += inflight_) {
@@ -200,33 +422,19 @@ class VerbsTransport final : public Transport {
         post_many(base, burst);
         flush();
       }
-      if (gpu) {
-        unsigned long long* d_bad = nullptr;
-        VB_THROW(hipMalloc((void**)&d_bad, 8) == hipSuccess, "hipMalloc");
-        hipMemset(d_bad, 0, 8);
-        VB_THROW(rocp2p_verify(region_gpu_, cfg_.region_bytes, seed, d_bad,
-                               0) == hipSuccess,
-                 "verify kernel");
-        unsigned long long h_bad = 0;
-        hipMemcpy(&h_bad, d_bad, 8, hipMemcpyDeviceToHost);
-        hipFree(d_bad);
-        return h_bad;
+      if (oob_) {
+        // REMOTE verification: the target audits its own region
+        oob_->send_kv({{"op", "verify"}, {"seed", std::to_string(seed)}});
+        return strtoull(oob_->recv_kv().at("bad").c_str(), nullptr, 10);
       }
-      const uint64_t* r = reinterpret_cast<const uint64_t*>(region_host_);
-      for (size_t w = 0; w < cfg_.region_bytes / 8; w++)
-        bad += (r[w] != rocp2p_pattern_word(seed, w));
-      return bad;
+      return region_.verify(seed);
     }
-    // read direction: pattern the region, RDMA_READ back, verify host side
-    if (gpu) {
-      VB_THROW(rocp2p_fill(region_gpu_, cfg_.region_bytes, seed, 0) ==
-                   hipSuccess,
-               "fill kernel");
-      hipDeviceSynchronize();
+    // read direction
+    if (oob_) {
+      oob_->send_kv({{"op", "fill"}, {"seed", std::to_string(seed)}});
+      oob_->recv_kv();
     } else {
-      uint64_t* r = reinterpret_cast<uint64_t*>(region_host_);
-      for (size_t w = 0; w < cfg_.region_bytes / 8; w++)
-        r[w] = rocp2p_pattern_word(seed, w);
+      region_.fill(seed);
     }
     for (size_t base = 0; base < msgs_per_region(); base += inflight_) {
       size_t burst = std::min(inflight_, msgs_per_region() - base);
@@ -243,78 +451,6 @@ class VerbsTransport final : public Transport {
   }
 
  private:
-  ibv_qp* make_qp() {
-    ibv_qp_init_attr a;
-    memset(&a, 0, sizeof(a));
-    a.send_cq = cq_;
-    a.recv_cq = cq_;
-    a.qp_type = IBV_QPT_RC;
-    a.cap.max_send_wr = (uint32_t)(2 * inflight_ + 8);
-    a.cap.max_recv_wr = 16;
-    a.cap.max_send_sge = 1;
-    a.cap.max_recv_sge = 1;
-    ibv_qp* qp = ibv_create_qp(pd_, &a);
-    VB_THROW(qp, "ibv_create_qp failed");
-    return qp;
-  }
-
-  void connect_loopback(ibv_qp* qp, ibv_qp* peer) {
-    ibv_port_attr pattr;
-    VB_THROW(ibv_query_port(ctx_, 1, &pattr) == 0, "ibv_query_port");
-
-    ibv_qp_attr at;
-    memset(&at, 0, sizeof(at));
-    at.qp_state = IBV_QPS_INIT;
-    at.pkey_index = 0;
-    at.port_num = 1;
-    at.qp_access_flags = IBV_ACCESS_LOCAL_WRITE | IBV_ACCESS_REMOTE_WRITE |
-                         IBV_ACCESS_REMOTE_READ;
-    VB_THROW(ibv_modify_qp(qp, &at,
-                           IBV_QP_STATE | IBV_QP_PKEY_INDEX | IBV_QP_PORT |
-                               IBV_QP_ACCESS_FLAGS) == 0,
-             "modify->INIT");
-
-    memset(&at, 0, sizeof(at));
-    at.qp_state = IBV_QPS_RTR;
-    at.path_mtu = pattr.active_mtu;
-    at.dest_qp_num = peer->qp_num;
-    at.rq_psn = 0;
-    at.max_dest_rd_atomic = 4;
-    at.min_rnr_timer = 12;
-    at.ah_attr.port_num = 1;
-    if (pattr.link_layer == IBV_LINK_LAYER_ETHERNET) {
-      union ibv_gid gid;
-      VB_THROW(ibv_query_gid(ctx_, 1, gid_index_, &gid) == 0,
-               "ibv_query_gid");
-      at.ah_attr.is_global = 1;
-      at.ah_attr.grh.dgid = gid;
-      at.ah_attr.grh.sgid_index = (uint8_t)gid_index_;
-      at.ah_attr.grh.hop_limit = 1;
-    } else {
-      at.ah_attr.is_global = 0;
-      at.ah_attr.dlid = pattr.lid;
-    }
-    VB_THROW(ibv_modify_qp(qp, &at,
-                           IBV_QP_STATE | IBV_QP_AV | IBV_QP_PATH_MTU |
-                               IBV_QP_DEST_QPN | IBV_QP_RQ_PSN |
-                               IBV_QP_MAX_DEST_RD_ATOMIC |
-                               IBV_QP_MIN_RNR_TIMER) == 0,
-             "modify->RTR");
-
-    memset(&at, 0, sizeof(at));
-    at.qp_state = IBV_QPS_RTS;
-    at.sq_psn = 0;
-    at.timeout = 14;
-    at.retry_cnt = 7;
-    at.rnr_retry = 7;
-    at.max_rd_atomic = 4;
-    VB_THROW(ibv_modify_qp(qp, &at,
-                           IBV_QP_STATE | IBV_QP_SQ_PSN | IBV_QP_TIMEOUT |
-                               IBV_QP_RETRY_CNT | IBV_QP_RNR_RETRY |
-                               IBV_QP_MAX_QP_RD_ATOMIC) == 0,
-             "modify->RTS");
-  }
-
   void drain(size_t at_least) {
     ibv_wc wc[16];
     size_t done = 0;
@@ -325,29 +461,95 @@ class VerbsTransport final : public Transport {
         VB_THROW(wc[i].status == IBV_WC_SUCCESS,
                  std::string("completion error: ") +
                      ibv_wc_status_str(wc[i].status));
-      done += n;
-      outstanding_ -= n;
+      done += (size_t)n;
+      outstanding_ -= (size_t)n;
     }
   }
 
+  OobSocket* oob_ = nullptr;
   ibv_context* ctx_ = nullptr;
   ibv_pd* pd_ = nullptr;
   ibv_cq* cq_ = nullptr;
-  ibv_qp* qp_send_ = nullptr;
-  ibv_qp* qp_recv_ = nullptr;
+  ibv_qp* qp_ = nullptr;
+  ibv_qp* qp_peer_ = nullptr;  // loopback passive end
   ibv_mr* staging_mr_ = nullptr;
-  ibv_mr* region_mr_ = nullptr;
   uint8_t* staging_ = nullptr;
-  uint8_t* region_host_ = nullptr;
-  uint8_t* region_gpu_ = nullptr;
-  uint8_t* region_ptr_ = nullptr;
-  std::string mr_mode_;
+  bool staging_hip_ = false;
+  Region region_;  // loopback only
+  uint64_t remote_addr_ = 0;
+  uint32_t remote_rkey_ = 0;
   size_t outstanding_ = 0;
-  int gid_index_ = 1;  // RoCEv2 default; override via env if needed
 };
 
 std::unique_ptr<Transport> make_verbs_transport(const TransportConfig& cfg) {
-  return std::make_unique<VerbsTransport>(cfg);
+  return std::make_unique<VerbsTransport>(cfg, nullptr);
+}
+
+std::unique_ptr<Transport> make_verbs_client(const TransportConfig& cfg,
+                                             const std::string& host,
+                                             int port) {
+  return std::make_unique<VerbsTransport>(cfg, oob_connect(host, port));
+}
+
+// Passive target: register the region, exchange QP/MR params, answer
+// verify/fill ops until "bye".  announce(port) fires once listening.
+int run_verbs_target(const TransportConfig& cfg, int port,
+                     void (*announce)(int)) {
+  ibv_context* ctx = open_first_device();
+  ibv_pd* pd = ibv_alloc_pd(ctx);
+  VB_THROW(pd, "ibv_alloc_pd failed");
+  ibv_cq* cq = ibv_create_cq(ctx, 64, nullptr, nullptr, 0);
+  VB_THROW(cq, "ibv_create_cq failed");
+  Region region;
+  region.create(pd, cfg.region_bytes, cfg.verbs_mr, cfg.device_index);
+  ibv_qp* qp = make_rc_qp(pd, cq, 64);
+
+  OobServer server(port);
+  if (announce) announce(server.port());
+  OobSocket* sock = server.accept_one();
+  KvMap cli = sock->recv_kv();
+  PeerInfo peer;
+  peer.qpn = (uint32_t)strtoul(cli.at("qpn").c_str(), nullptr, 10);
+  peer.lid = (uint16_t)strtoul(cli.at("lid").c_str(), nullptr, 10);
+  gid_unhex(cli.at("gid"), peer.gid);
+  peer.mtu = atoi(cli.at("mtu").c_str());
+
+  PeerInfo mine = local_info(ctx, qp, env_gid_index());
+  sock->send_kv({{"qpn", std::to_string(mine.qpn)},
+                 {"lid", std::to_string(mine.lid)},
+                 {"gid", gid_hex(mine.gid)},
+                 {"mtu", std::to_string(mine.mtu)},
+                 {"raddr", std::to_string((uint64_t)region.ptr)},
+                 {"rkey", std::to_string(region.mr->rkey)},
+                 {"rbytes", std::to_string(region.bytes)}});
+  qp_to_rts(ctx, qp, peer, env_gid_index());
+
+  // control loop: the data plane never touches this thread
+  try {
+    for (;;) {
+      KvMap msg = sock->recv_kv();
+      auto op = msg.count("op") ? msg.at("op") : "bye";
+      if (op == "verify") {
+        uint64_t seed = strtoull(msg.at("seed").c_str(), nullptr, 10);
+        sock->send_kv({{"bad", std::to_string(region.verify(seed))}});
+      } else if (op == "fill") {
+        uint64_t seed = strtoull(msg.at("seed").c_str(), nullptr, 10);
+        region.fill(seed);
+        sock->send_kv({{"ok", "1"}});
+      } else {
+        break;
+      }
+    }
+  } catch (const std::exception&) {
+    // client went away: normal shutdown
+  }
+  delete sock;
+  ibv_destroy_qp(qp);
+  region.destroy();
+  ibv_destroy_cq(cq);
+  ibv_dealloc_pd(pd);
+  ibv_close_device(ctx);
+  return 0;
 }
 
 }  // namespace rocp2p
@@ -360,6 +562,15 @@ std::unique_ptr<Transport> make_verbs_transport(const TransportConfig&) {
   throw std::runtime_error(
       "verbs backend compiled out: <infiniband/verbs.h> (rdma-core) was "
       "not present at build time. Rebuild on an HCA-equipped host.");
+}
+
+std::unique_ptr<Transport> make_verbs_client(const TransportConfig&,
+                                             const std::string&, int) {
+  throw std::runtime_error("verbs backend compiled out (see above)");
+}
+
+int run_verbs_target(const TransportConfig&, int, void (*)(int)) {
+  throw std::runtime_error("verbs backend compiled out (see above)");
 }
 
 }  // namespace rocp2p

@@ -118,3 +118,32 @@ def test_verbs_backend_lat_against_fake_layer(built_fakeverbs):
     assert out.returncode == 0, out.stderr
     r = json.loads(out.stdout.strip())
     assert r["mode"] == "lat" and r["integrity"] == "ok"
+
+
+@pytest.mark.parametrize("link", ["ib", "eth"])
+@pytest.mark.timeout(300)
+def test_verbs_client_server_remote_selftest(built_fakeverbs, link):
+    """Full ib_write_bw client/server shape through the real verbs
+    backend: TCP OOB exchange, remote QP bring-up, one-sided writes
+    into the server's registered region, REMOTE integrity verdict."""
+    env = dict(os.environ, FAKE_VERBS_LINK=link)
+    out = subprocess.run(
+        [built_fakeverbs, "--transport", "verbs", "--mr", "host",
+         "--msg", "65536", "--region", "2097152", "--remote-selftest",
+         "--secs", "0.1"],
+        capture_output=True, text=True, timeout=120, env=env)
+    assert out.returncode == 0, out.stderr
+    r = json.loads(out.stdout.strip())
+    assert r["mode"] == "remote-selftest"
+    assert r["remote_integrity"] == "ok"
+
+
+@pytest.mark.timeout(300)
+def test_verbs_client_server_read_direction(built_fakeverbs):
+    out = subprocess.run(
+        [built_fakeverbs, "--transport", "verbs", "--mr", "host",
+         "--msg", "4096", "--region", "1048576", "--remote-selftest",
+         "--secs", "0.1", "--dir", "read"],
+        capture_output=True, text=True, timeout=120)
+    assert out.returncode == 0, out.stderr
+    assert json.loads(out.stdout.strip())["remote_integrity"] == "ok"
