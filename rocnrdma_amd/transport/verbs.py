@@ -1,8 +1,12 @@
-"""Real IB-verbs transport (PeerDirect path) — drives the C++ harness
-(harness/) through its shared library when libibverbs and an HCA are
-present.  On this pool no HCA/libibverbs exists, so this module degrades
-to a capability probe; the full loopback QP wiring lives in
-harness/src/verbs_backend.cpp and activates on verbs-equipped hosts.
+"""IB-verbs capability probe (PeerDirect path).
+
+The verbs data plane lives in the native harness
+(harness/src/verbs_backend.cpp): loopback QP pair, client/server mode
+with TCP OOB bootstrap, peer/dmabuf/host MR modes — CI-executed through
+the fake-verbs layer and activated for real on HCA-equipped hosts.
+This module only answers "is a usable verbs stack present?" (dlopen +
+device count) so auto-transport selection and the runbook's layer
+isolation work; it deliberately implements no python data plane.
 """
 from __future__ import annotations
 
