@@ -156,12 +156,17 @@ int main(int argc, char** argv) {
   if (serve_port >= 0) {
     // passive target (ib_write_bw server shape); blocks until the
     // client says bye
-    run_verbs_target(cfg, serve_port,
-                     [](int p) { printf("listening on port %d\n", p); });
+    try {
+      run_verbs_target(cfg, serve_port,
+                       [](int p) { printf("listening on port %d\n", p); });
+    } catch (const std::exception& e) {
+      fprintf(stderr, "serve: %s\n", e.what());
+      return 3;
+    }
     return 0;
   }
 
-  if (remote_selftest || !connect_to.empty()) {
+  if (remote_selftest || !connect_to.empty()) try {
     // client/server data plane through the verbs backend.  selftest:
     // both endpoints in THIS process over real TCP loopback (runs
     // under the fake-verbs CI layer; on an HCA host it exercises the
@@ -207,6 +212,9 @@ int main(int argc, char** argv) {
     tp.reset();  // sends nothing; dtor closes OOB -> server exits loop
     if (server_thread.joinable()) server_thread.join();
     return (integrity && r.bad) ? 1 : 0;
+  } catch (const std::exception& e) {
+    fprintf(stderr, "client: %s\n", e.what());
+    return 3;
   }
 
   std::vector<size_t> sizes =
