@@ -54,3 +54,42 @@ def test_remote_verify_detects_wrong_seed(target):
     tp.oob.send({"op": "verify", "seed": 6})
     assert tp.oob.recv()["bad"] > 0
     tp.close()
+
+
+@pytest.mark.timeout(180)
+def test_remote_cli_end_to_end():
+    """The harness.remote CLI: --serve in one process, --connect in
+    another, remote verification verdict in the client output."""
+    import os
+    import re
+    import subprocess
+    import sys
+    import time
+
+    ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ)
+    env["PYTHONPATH"] = ROOT + os.pathsep + env.get("PYTHONPATH", "")
+    srv = subprocess.Popen(
+        [sys.executable, "-m", "rocnrdma_amd.harness.remote", "--serve",
+         "--region", str(REGION)],
+        stdout=subprocess.PIPE, stderr=subprocess.PIPE, text=True,
+        env=env, cwd=ROOT)
+    try:
+        line = srv.stdout.readline()
+        m = re.search(r"OOB port (\d+)", line)
+        assert m, line
+        port = m.group(1)
+        out = subprocess.run(
+            [sys.executable, "-m", "rocnrdma_amd.harness.remote",
+             "--connect", f"127.0.0.1:{port}", "--region", str(REGION),
+             "--msg", str(MSG), "--secs", "0.3"],
+            capture_output=True, text=True, env=env, cwd=ROOT,
+            timeout=120)
+        assert out.returncode == 0, out.stderr
+        assert "'remote_verify_bad': 0" in out.stdout
+    finally:
+        deadline = time.time() + 20
+        while srv.poll() is None and time.time() < deadline:
+            time.sleep(0.2)
+        if srv.poll() is None:
+            srv.kill()
