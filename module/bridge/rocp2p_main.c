@@ -264,6 +264,13 @@ static int rocp2p_get_pages(unsigned long addr, size_t size, int write,
 	}
 	reg->core_context = core_context;
 
+	/* This KFD call is made UNDER reg->lock — the exception to the
+	 * calls-outside-the-lock rule (see put_pages): no pin exists for
+	 * this reg yet, so no free callback can be in flight for it, and
+	 * holding the lock closes the install race (a revoke firing right
+	 * after KFD pins blocks on the lock until state=PINNED below, then
+	 * tears down normally).  Requirement on KFD: it must not invoke
+	 * this pin's free callback synchronously from inside get_pages. */
 	ret = rdma_interface->get_pages(addr, size, reg->pid,
 #if ROCNR_AMD_RDMA_HAS_DMA_DEV
 					NULL,	/* mapping done per-HCA in dma_map */
