@@ -15,12 +15,10 @@
  *
  *   ROCNR_PEER_MEM_CORE_CONTEXT_U64   (default 1)
  *       MLNX_OFED >= 4.0 passes the IB core's registration cookie to
- *       get_pages as a u64; the 2016 ABI used void*.
- *
- *   ROCNR_PEER_MEM_HAVE_SYSTEM        (default 0)
- *       Newer OFED peer_mem clients may expose PEER_MEM_INVALIDATE_UNMAPS
- *       style flags; we keep the core seven-callback surface and gate
- *       extras here.
+ *       get_pages as a u64; the 2016 ABI used void*.  Newer OFED also
+ *       added optional capability flags (e.g. PEER_MEM_INVALIDATE_
+ *       UNMAPS); the seven-callback surface below is the stable core
+ *       every version dispatches through.
  *
  * When building against a real OFED tree (OFA_DIR set), the module Makefile
  * defines ROCNR_USE_SYSTEM_PEER_MEM and this header simply includes the
