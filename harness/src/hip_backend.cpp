@@ -42,7 +42,7 @@ class HipTransport final : public Transport {
     if (cfg.msg_bytes % 16)
       throw std::runtime_error("msg size must be a 16-byte multiple");
     engine_kernel_ = cfg.engine == "kernel" ||
-                     (cfg.engine == "auto" && cfg.msg_bytes < (4u << 20));
+                     (cfg.engine == "auto" && cfg.msg_bytes < (8u << 20));
     HIP_THROW(hipSetDevice(cfg.device_index));
 
     inflight_ = cfg.inflight;

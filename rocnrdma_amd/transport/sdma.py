@@ -17,7 +17,7 @@ Two engines, mirroring how a NIC actually retires work:
   host-pinned staging directly over PCIe (fine-grained zero-copy).
   Small-message bandwidth becomes PCIe-bound instead of launch-bound.
 
-"auto" picks kernel below 4 MiB messages.  Integrity is proven on-GPU
+"auto" picks kernel below 8 MiB messages (measured crossover).  Integrity is proven on-GPU
 (verify/CRC kernels) — zero host readback for write direction.
 """
 from __future__ import annotations
@@ -27,7 +27,7 @@ import torch
 from ..utils import pattern
 from .base import Transport
 
-_KERNEL_THRESHOLD = 4 << 20
+_KERNEL_THRESHOLD = 8 << 20
 _STAGING_TARGET = 64 << 20  # pinned staging budget for small messages
 
 
