@@ -9,7 +9,7 @@ One transport instance = one "QP" analog on one GPU.
 Two engines, mirroring how a NIC actually retires work:
 
 - "stream": hipMemcpyAsync per message on round-robin HIP streams (SDMA
-  engines).  ~10 us host cost per message — fine >= 4 MiB, hopeless at
+  engines).  ~10 us host cost per message — fine >= 8 MiB, hopeless at
   4 KiB (measured 0.3 GB/s).
 - "kernel": doorbell semantics.  post() appends a WQE (two u64 writes
   into a pinned descriptor ring); flush() rings the doorbell — one
