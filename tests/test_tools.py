@@ -63,3 +63,52 @@ def test_run_matrix_quick_cpu():
 
     r = json.loads(out.stdout)
     assert r["configs"]["1_host_loopback"]["integrity_bad"] == 0
+
+
+MODERN_PEER = """
+struct peer_memory_client {
+  int (*get_pages)(unsigned long addr, size_t size, int write, int force,
+                   struct sg_table *sg_head, void *client_context,
+                   u64 core_context);
+};
+"""
+LEGACY_PEER = """
+struct peer_memory_client {
+  int (*get_pages)(unsigned long addr, size_t size, int write, int force,
+                   struct sg_table *sg_head, void *client_context,
+                   void *core_context);
+};
+"""
+MODERN_AMDR = """
+struct amd_rdma_interface {
+  int (*get_pages)(uint64_t address, uint64_t length, struct pid *pid,
+                   struct device *dma_dev,
+                   struct amd_p2p_info **amd_p2p_data,
+                   void (*free_callback)(void *), void *client_priv);
+};
+"""
+LEGACY_AMDR = """
+struct amd_rdma_interface {
+  int (*get_pages)(uint64_t address, uint64_t length, struct pid *pid,
+                   struct amd_p2p_info **amd_p2p_data,
+                   void (*free_callback)(void *), void *client_priv);
+};
+"""
+
+
+@pytest.mark.parametrize("peer,amdr,u64,dmadev", [
+    (MODERN_PEER, MODERN_AMDR, 1, 1),
+    (LEGACY_PEER, LEGACY_AMDR, 0, 0),
+    (MODERN_PEER, LEGACY_AMDR, 1, 0),
+])
+def test_abi_probe_classifies(tmp_path, peer, amdr, u64, dmadev):
+    p = tmp_path / "peer_mem.h"
+    a = tmp_path / "amd_rdma.h"
+    p.write_text(peer)
+    a.write_text(amdr)
+    out = subprocess.run(
+        ["sh", os.path.join(ROOT, "tools", "abi_probe.sh"), str(p), str(a)],
+        capture_output=True, text=True, timeout=60)
+    assert out.returncode == 0, out.stdout + out.stderr
+    assert f"ROCNR_PEER_MEM_CORE_CONTEXT_U64={u64}" in out.stdout
+    assert f"ROCNR_AMD_RDMA_HAS_DMA_DEV={dmadev}" in out.stdout
