@@ -14,6 +14,7 @@
 
 #include <assert.h>
 #include <time.h>
+#include <stdlib.h>
 #include <unistd.h>
 
 #define CHECK(cond) do { \
@@ -312,7 +313,11 @@ static void *race_free_thread(void *argp)
 
 static void test_invalidate_race(void)
 {
-	enum { ITERS = 3000 };
+	int ITERS = 3000;
+	const char *env_iters = getenv("ROCNR_RACE_ITERS");
+
+	if (env_iters && atoi(env_iters) > 0)
+		ITERS = atoi(env_iters);
 	unsigned int seed = 12345;
 	int i;
 

@@ -13,6 +13,7 @@
 
 #include <assert.h>
 #include <time.h>
+#include <stdlib.h>
 
 #define CHECK(cond) do { \
 	if (!(cond)) { \
@@ -179,7 +180,11 @@ static void *race_free_thread(void *argp)
 
 static void test_revoke_vs_put_race(void)
 {
-	enum { ITERS = 2000 };
+	int ITERS = 2000;
+	const char *env_iters = getenv("ROCNR_RACE_ITERS");
+
+	if (env_iters && atoi(env_iters) > 0)
+		ITERS = atoi(env_iters);
 	unsigned int seed = 777;
 	int i;
 

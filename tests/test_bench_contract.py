@@ -80,3 +80,9 @@ def test_two_rank_gloo_aggregate():
     r = json.loads(lines[0])
     assert r["n_gpus"] == 2
     assert "x2" in r["config"]["parallelism"]
+
+
+def test_msgs_per_step_override():
+    r = run_bench(["--msgs-per-step", "4"])
+    assert r["config"]["global_batch"] == 4
+    assert r["value"] > 0
