@@ -139,8 +139,31 @@ def main():
         bad = tp.integrity_check(seed=0xC0FFEE + rank)
         integrity = "ok" if bad == 0 else f"FAILED:{bad}"
 
+    # the metric names 4KB/1MB/64MB: attach quick per-GPU points for the
+    # other two sizes (outside the timed region; single-rank runs only —
+    # the scaling runs keep the headline size)
+    msg_sweep = None
+    if world == 1 and tp.name in ("sdma", "fake") and not args.msgs_per_step:
+        from rocnrdma_amd.harness.sweep import run_point
+        from rocnrdma_amd.transport import get_transport as _gt
+
+        msg_sweep = {}
+        for other in (4096, 1 << 20):
+            if other == msg:
+                continue
+            stp = _gt(tp.name, msg_bytes=other,
+                      region_bytes=max(region // other, 1) * other,
+                      direction=args.direction, device=device)
+            try:
+                msg_sweep[str(other)] = run_point(
+                    stp, target_secs=0.3)["gbps"]
+            finally:
+                stp.close()
+
     total_bytes = bytes_per_step * args.steps * world
     value = total_bytes / elapsed / 1e9
+    if msg_sweep is not None:
+        msg_sweep[str(msg)] = round(value, 3)
 
     result = {
         "metric": METRIC,
@@ -168,6 +191,7 @@ def main():
             "streams": args.streams,
             "integrity": integrity,
             "numa_bound": numa,
+            "msg_sweep_gbps": msg_sweep,
         },
     }
     if rank == 0:

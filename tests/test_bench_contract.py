@@ -86,3 +86,11 @@ def test_msgs_per_step_override():
     r = run_bench(["--msgs-per-step", "4"])
     assert r["config"]["global_batch"] == 4
     assert r["value"] > 0
+
+
+def test_msg_sweep_embedded_single_rank():
+    r = run_bench([])
+    sw = r["config"]["msg_sweep_gbps"]
+    assert sw is not None
+    assert str(r["config"]["msg_bytes"]) in sw
+    assert "4096" in sw and all(v > 0 for v in sw.values())
