@@ -58,3 +58,10 @@ def test_abi_matches_c_header(tmp_path):
     assert sizes == [24, 16, 24, 16, 48]
     assert codes == [abi.GET_PAGE_SIZE, abi.GET_PAGES, abi.PUT_PAGES,
                      abi.IS_GPU_ADDRESS, abi.GET_INFO]
+
+
+def test_device_path_matches_header():
+    hdr = open(os.path.join(ROOT, "module", "include",
+                            "rocp2p_probe_abi.h")).read()
+    assert f'"{abi.DEVICE_PATH}"' in hdr
+    assert f"'{chr(abi.MAGIC)}'" in hdr

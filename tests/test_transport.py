@@ -85,3 +85,13 @@ def test_metrics_capture_in_soak():
     assert "rocp2p_bytes_total" in text
     assert "rocp2p_integrity_audits_total" in text
     assert stats["failures"] == 0
+
+
+def test_available_transports_cpu_box():
+    names = available_transports()
+    assert "fake" in names and "shm" in names
+    # no GPU in CI: sdma must NOT be offered
+    import torch
+
+    if not torch.cuda.is_available():
+        assert "sdma" not in names
