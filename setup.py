@@ -4,7 +4,7 @@
 # rocnrdma_amd/ops/ and travels to the GPU box with the repo snapshot)
 import os
 
-from setuptools import setup
+from setuptools import find_packages, setup
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
@@ -13,7 +13,7 @@ from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E40
 setup(
     name="rocnrdma_amd",
     version="0.1.0",
-    packages=["rocnrdma_amd"],
+    packages=find_packages(include=["rocnrdma_amd", "rocnrdma_amd.*"]),
     ext_modules=[
         CUDAExtension(
             name="rocnrdma_amd.ops._p2p_ext",
