@@ -179,17 +179,30 @@ int main(int argc, char** argv) {
       static std::condition_variable s_cv;
       TransportConfig scfg = cfg;
       server_thread = std::thread([scfg] {
-        run_verbs_target(scfg, 0, [](int p) {
+        try {
+          run_verbs_target(scfg, 0, [](int p) {
+            {
+              std::lock_guard<std::mutex> g(s_mu);
+              s_port.store(p);
+            }
+            s_cv.notify_all();
+          });
+        } catch (const std::exception& e) {
+          fprintf(stderr, "selftest server: %s\n", e.what());
           {
             std::lock_guard<std::mutex> g(s_mu);
-            s_port.store(p);
+            s_port.store(-1);  // unblock the client side
           }
           s_cv.notify_all();
-        });
+        }
       });
       {
         std::unique_lock<std::mutex> lk(s_mu);
         s_cv.wait(lk, [] { return s_port.load() != 0; });
+      }
+      if (s_port.load() < 0) {
+        server_thread.join();
+        throw std::runtime_error("selftest server failed to start");
       }
       tp = make_verbs_client(cfg, "127.0.0.1", s_port.load());
     } else {

@@ -66,6 +66,7 @@ bool verbs_runtime_available() {
 #ifdef ROCP2P_HAVE_VERBS
 #include <infiniband/verbs.h>
 #include <hip/hip_runtime.h>
+#include <unistd.h>
 
 #include <cstdio>
 #include <vector>
@@ -238,6 +239,7 @@ struct Region {
                    hipSuccess,
                "hipMemGetHandleForAddressRange(dmabuf) failed");
       mr = ibv_reg_dmabuf_mr(pd, 0, nbytes, (uint64_t)gpu, fd, acc);
+      close(fd);  // the MR holds its own reference to the dmabuf
       VB_THROW(mr, "ibv_reg_dmabuf_mr failed");
     } else {
       // peer mode: dispatched to the rocp2p bridge by the IB core
