@@ -41,9 +41,13 @@ def target_serve(port_conn, region_bytes: int, port: int = 0) -> None:
     try:
         server.accept()
         server.send({"shm_name": shm.name, "region_bytes": region_bytes})
-        # passive until the initiator asks for verification or bye
+        # passive until the initiator asks for verification or bye;
+        # an initiator crash (socket close) is a normal shutdown
         while True:
-            msg = server.recv()
+            try:
+                msg = server.recv()
+            except (ConnectionError, OSError):
+                break
             if msg.get("op") == "verify":
                 ref = pattern.fill_reference(region_bytes, msg["seed"])
                 got = np.frombuffer(shm.buf, dtype=np.uint8)
