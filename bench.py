@@ -115,6 +115,23 @@ def main():
         tp.post_many(base, msgs_per_step)
         tp.flush()
 
+    # synthetic random payloads: pattern every staging slot before the
+    # timed region (PCIe is data-agnostic, but the measurement should
+    # not depend on that)
+    if hasattr(tp, "staging"):
+        from rocnrdma_amd.utils import pattern as _pat
+
+        for si, slot in enumerate(tp.staging):
+            words = _pat.splitmix64_words(0xDA7A + rank, si * 8192,
+                                          msg // 8)
+            try:
+                import numpy as _np
+
+                slot_np = slot.numpy() if hasattr(slot, "numpy") else slot
+                slot_np.view(_np.uint64)[:] = words
+            except (TypeError, ValueError):
+                break
+
     # warmup (also first-touch of staging and region)
     for s in range(args.warmup):
         step(s * msgs_per_step)
