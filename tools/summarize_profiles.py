@@ -1,0 +1,64 @@
+#!/usr/bin/env python3
+"""One-screen summary of the committed measurement artifacts
+(profiles/*.json) — the quick 'what did we measure' view.
+
+Usage: python tools/summarize_profiles.py
+"""
+from __future__ import annotations
+
+import glob
+import json
+import os
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+P = os.path.join(ROOT, "profiles")
+
+
+def load(name):
+    path = os.path.join(P, name)
+    if not os.path.exists(path):
+        return None
+    with open(path) as f:
+        return json.load(f)
+
+
+def main():
+    b = load("bench_n1.json")
+    if b:
+        print(f"headline bench      : {b['value']} {b['unit']} "
+              f"({b['config']['msg_bytes']} B msgs, "
+              f"integrity {b['config']['integrity']})")
+        sweep = b["config"].get("msg_sweep_gbps")
+        if sweep:
+            pts = ", ".join(f"{int(k) >> 10} KiB={v}" for k, v in
+                            sorted(sweep.items(), key=lambda kv: int(kv[0])))
+            print(f"  msg points        : {pts}")
+    s = load("sweep_1gpu.json")
+    if s:
+        for row in s:
+            if row["direction"] == "write":
+                print(f"sweep write {row['msg_bytes']:>9} B : "
+                      f"{row['gbps']:7.2f} GB/s  ({row['mops']:.3f} Mmsg/s)")
+    d = load("devbw_2gib.json")
+    if d:
+        print("device kernels (2 GiB): " + ", ".join(
+            f"{k.split('_')[0]}={v:.0f}" for k, v in d.items()
+            if k.endswith("GBps") or k.endswith("GBps_rw")))
+    d = load("devbw_240gib.json")
+    if d:
+        print("device kernels (240 GiB footprint): " + ", ".join(
+            f"{k.split('_')[0]}={v:.0f}" for k, v in d.items()
+            if "GBps" in k))
+    for f in sorted(glob.glob(os.path.join(P, "soak_*.txt"))):
+        with open(f) as fh:
+            print(os.path.basename(f), "->", fh.read().strip()[:100])
+    m = load("config_matrix_1gpu.json")
+    if m:
+        done = [k for k, v in m["configs"].items()
+                if not isinstance(v, str)]
+        print(f"config matrix       : {len(done)} configs measured "
+              f"({', '.join(done)})")
+
+
+if __name__ == "__main__":
+    main()
