@@ -129,9 +129,23 @@ static void rocp2p_check_bar_aperture(void)
 
 enum rocp2p_state {
 	ROCP2P_ACQUIRED = 0,	/* context exists, nothing pinned */
+	ROCP2P_PINNING,		/* KFD get_pages in flight (lock dropped) */
 	ROCP2P_PINNED,		/* KFD holds a pin for us */
 	ROCP2P_REVOKED,		/* KFD invalidated; pin is gone */
 };
+
+/* Bring-up fallback for amd_rdma ABI drift: the modern ROCK
+ * get_pages takes the DMA target device and may refuse dma_dev=NULL (it
+ * dma-maps against that device internally).  When enabled (default) and
+ * KFD rejects the NULL-device pin, the pin is DEFERRED to the first
+ * dma_map call, where the requesting HCA's device is known; that pin's
+ * sg table is then used as-is (KFD already device-mapped it) and
+ * dma_map_resource is skipped — the reference-style single-device path
+ * (amdp2p.c:222-240).  Additional HCAs mapping the same MR reuse that
+ * table with a loud warning (correct only when the IOMMU does not
+ * isolate the two HCAs differently).  See docs/LIMITATIONS.md. */
+static bool null_dev_fallback = true;
+module_param(null_dev_fallback, bool, 0444);
 
 /* One dma_map() result for one device. */
 struct rocp2p_dmamap {
@@ -151,6 +165,8 @@ struct rocp2p_reg {
 	enum rocp2p_state state;
 	struct amd_p2p_info *p2p;
 	unsigned long page_size;	/* cached from KFD */
+	bool defer_pin;		/* KFD refused dma_dev=NULL: pin at dma_map */
+	struct device *pin_dev;	/* device the deferred pin was made for */
 	struct list_head dmamaps;
 
 	rocnr_core_context_t core_context;
@@ -171,6 +187,19 @@ static void rocp2p_kfd_free_cb(void *client_priv)
 	}
 
 	mutex_lock(&reg->lock);
+	if (reg->state == ROCP2P_PINNING) {
+		/* Revoke raced the pin install (possibly synchronously from
+		 * inside KFD's get_pages).  Mark revoked and do NOT upcall:
+		 * no MR exists yet — the in-flight get_pages/dma_map sees
+		 * REVOKED, discards the (KFD-reclaimed) pin without touching
+		 * it, and fails the registration; the IB core unwinds via
+		 * release(). */
+		reg->state = ROCP2P_REVOKED;
+		mutex_unlock(&reg->lock);
+		atomic64_inc(&rocp2p_invalidations);
+		rp_dbg("invalidate during pin install\n");
+		return;
+	}
 	if (reg->state != ROCP2P_PINNED) {
 		/* Duplicate or late revoke: nothing to tear down. */
 		mutex_unlock(&reg->lock);
@@ -239,14 +268,74 @@ static int rocp2p_acquire(unsigned long addr, size_t size,
 	return 1;
 }
 
+/* Pin reg's range through KFD.  Lock-free install (probe module's
+ * PIN_INIT pattern): the KFD call is made with reg->lock
+ * DROPPED and state=PINNING, so a revoke firing during the pin — even
+ * synchronously from inside KFD's get_pages — takes the lock, finds
+ * PINNING, marks REVOKED and returns; we then discard the KFD-reclaimed
+ * pin without dereferencing it.  dma_dev is NULL for the pin-at-
+ * registration path and the HCA's device for the deferred drift
+ * fallback (see null_dev_fallback). */
+static int rocp2p_pin(struct rocp2p_reg *reg, struct device *dma_dev)
+{
+	struct amd_p2p_info *p2p = NULL;
+	unsigned long page_size = 0;
+	int ret;
+
+	mutex_lock(&reg->lock);
+	if (reg->state != ROCP2P_ACQUIRED) {
+		mutex_unlock(&reg->lock);
+		return -EINVAL;
+	}
+	reg->state = ROCP2P_PINNING;
+	mutex_unlock(&reg->lock);
+
+	ret = rdma_interface->get_pages(reg->va, reg->size, reg->pid,
+#if ROCNR_AMD_RDMA_HAS_DMA_DEV
+					dma_dev,
+#endif
+					&p2p, rocp2p_kfd_free_cb, reg);
+#if !ROCNR_AMD_RDMA_HAS_DMA_DEV
+	(void)dma_dev;
+#endif
+
+	if (!ret && p2p &&
+	    (rdma_interface->get_page_size(reg->va, reg->size, reg->pid,
+					   &page_size) || !page_size)) {
+		/* MI355X VRAM granule; only used for reporting. */
+		page_size = 2UL << 20;
+		rp_warn("get_pages: page-size query failed, assuming 2 MiB\n");
+	}
+
+	mutex_lock(&reg->lock);
+	if (reg->state == ROCP2P_REVOKED) {
+		/* Revoke won the race: KFD reclaimed the pin when the free
+		 * callback returned — p2p (if any) is dead memory and must
+		 * not be touched or put. */
+		mutex_unlock(&reg->lock);
+		rp_warn("pin: revoked during install\n");
+		return -ENODEV;
+	}
+	if (ret || !p2p) {
+		reg->state = ROCP2P_ACQUIRED;
+		mutex_unlock(&reg->lock);
+		return ret ? ret : -ENOMEM;
+	}
+	reg->p2p = p2p;
+	reg->page_size = page_size;
+	reg->pin_dev = dma_dev;
+	reg->state = ROCP2P_PINNED;
+	mutex_unlock(&reg->lock);
+	atomic64_add(reg->size, &rocp2p_pinned_bytes);
+	return 0;
+}
+
 static int rocp2p_get_pages(unsigned long addr, size_t size, int write,
 			    int force, struct sg_table *sg_head,
 			    void *client_context,
 			    rocnr_core_context_t core_context)
 {
 	struct rocp2p_reg *reg = client_context;
-	struct amd_p2p_info *p2p = NULL;
-	unsigned long page_size = 0;
 	int ret;
 
 	if (!reg)
@@ -258,47 +347,29 @@ static int rocp2p_get_pages(unsigned long addr, size_t size, int write,
 	}
 
 	mutex_lock(&reg->lock);
-	if (reg->state != ROCP2P_ACQUIRED) {
-		mutex_unlock(&reg->lock);
-		return -EINVAL;
-	}
 	reg->core_context = core_context;
-
-	/* This KFD call is made UNDER reg->lock — the exception to the
-	 * calls-outside-the-lock rule (see put_pages): no pin exists for
-	 * this reg yet, so no free callback can be in flight for it, and
-	 * holding the lock closes the install race (a revoke firing right
-	 * after KFD pins blocks on the lock until state=PINNED below, then
-	 * tears down normally).  Requirement on KFD: it must not invoke
-	 * this pin's free callback synchronously from inside get_pages. */
-	ret = rdma_interface->get_pages(addr, size, reg->pid,
-#if ROCNR_AMD_RDMA_HAS_DMA_DEV
-					NULL,	/* mapping done per-HCA in dma_map */
-#endif
-					&p2p, rocp2p_kfd_free_cb, reg);
-	if (ret || !p2p) {
-		mutex_unlock(&reg->lock);
-		rp_err("get_pages: KFD pin failed: %d\n", ret);
-		return ret ? ret : -ENOMEM;
-	}
-
-	if (rdma_interface->get_page_size(addr, size, reg->pid, &page_size) ||
-	    !page_size) {
-		/* MI355X VRAM granule; only used for reporting. */
-		page_size = 2UL << 20;
-		rp_warn("get_pages: page-size query failed, assuming 2 MiB\n");
-	}
-
-	reg->p2p = p2p;
-	reg->page_size = page_size;
-	reg->state = ROCP2P_PINNED;
 	mutex_unlock(&reg->lock);
 
-	atomic64_add(size, &rocp2p_pinned_bytes);
+	ret = rocp2p_pin(reg, NULL);
+#if ROCNR_AMD_RDMA_HAS_DMA_DEV
+	if (ret && ret != -ENODEV && null_dev_fallback) {
+		/* Drift fallback: modern ROCK KFD may require the DMA
+		 * target device (it maps against it inside get_pages).
+		 * Defer the pin to dma_map, where the HCA is known. */
+		rp_warn("get_pages: KFD pin with dma_dev=NULL failed (%d); deferring pin to dma_map (amd_rdma ABI drift fallback — see docs/LIMITATIONS.md)\n",
+			ret);
+		mutex_lock(&reg->lock);
+		reg->defer_pin = true;
+		mutex_unlock(&reg->lock);
+		return 0;
+	}
+#endif
+	if (ret)
+		rp_err("get_pages: KFD pin failed: %d\n", ret);
 	/* sg_head intentionally not filled here: translation happens in
 	 * dma_map where the target device is known (reference kept the
 	 * same deferral — amdp2p.c:214). */
-	return 0;
+	return ret;
 }
 
 /* Iterate KFD's pinned sg table as bus-address segments. */
@@ -344,6 +415,18 @@ static int emit_map_one(void *ctx, const struct rocnr_seg *seg)
 	return 0;
 }
 
+/* Deferred-pin path: KFD already device-mapped the sg table inside
+ * get_pages(dma_dev); copy the coalesced runs verbatim. */
+static int emit_copy_one(void *ctx, const struct rocnr_seg *seg)
+{
+	struct emit_map_ctx *e = ctx;
+
+	sg_dma_address(e->sg) = (dma_addr_t)seg->addr;
+	sg_dma_len(e->sg) = seg->len;
+	e->sg = sg_next(e->sg);
+	return 0;
+}
+
 static void rocp2p_unmap_one(struct rocp2p_dmamap *map, size_t upto)
 {
 	struct scatterlist *sg;
@@ -372,12 +455,28 @@ static int rocp2p_dma_map(struct sg_table *sg_head, void *client_context,
 	struct emit_map_ctx emit;
 	u64 max_seg;
 	size_t nsegs, nout;
+	bool deferred;
 	int ret;
 
 	if (!reg || !dma_device || !nmap)
 		return -EINVAL;
 
 	mutex_lock(&reg->lock);
+	if (reg->defer_pin && reg->state == ROCP2P_ACQUIRED) {
+		/* Drift fallback: execute the deferred pin now, against the
+		 * requesting HCA's device (see null_dev_fallback). */
+		mutex_unlock(&reg->lock);
+		ret = rocp2p_pin(reg, dma_device);
+		if (ret) {
+			rp_err("dma_map: deferred KFD pin failed: %d\n", ret);
+			return ret;
+		}
+		mutex_lock(&reg->lock);
+	}
+	deferred = reg->defer_pin;
+	if (deferred && reg->state == ROCP2P_PINNED &&
+	    reg->pin_dev != dma_device)
+		rp_warn("dma_map: deferred pin was device-mapped for another HCA; sharing its table (reference-style path — IOMMU isolation between HCAs is NOT honored)\n");
 	if (reg->state != ROCP2P_PINNED || !reg->p2p || !reg->p2p->pages) {
 		mutex_unlock(&reg->lock);
 		rp_err("dma_map: no pinned pages (state %d)\n", reg->state);
@@ -406,7 +505,7 @@ static int rocp2p_dma_map(struct sg_table *sg_head, void *client_context,
 		return ret;
 	}
 	map->dev = dma_device;
-	map->mapped = true;
+	map->mapped = !deferred;
 	list_add(&map->node, &reg->dmamaps);
 
 	emit.dev = dma_device;
@@ -414,7 +513,8 @@ static int rocp2p_dma_map(struct sg_table *sg_head, void *client_context,
 	emit.mapped = 0;
 	emit.err = 0;
 	kit.sg = reg->p2p->pages->sgl;
-	nout = rocnr_coalesce(&kit.it, max_seg, emit_map_one, &emit);
+	nout = rocnr_coalesce(&kit.it, max_seg,
+			      deferred ? emit_copy_one : emit_map_one, &emit);
 	if (nout == (size_t)-1 || emit.err) {
 		size_t done = emit.mapped;
 
@@ -593,6 +693,14 @@ static int __init rocp2p_init(void)
 	ib_reg_handle = ib_register_peer_memory_client(&rocp2p_client,
 						       &ib_invalidate_cb);
 	if (!ib_reg_handle || !ib_invalidate_cb) {
+		if (ib_reg_handle) {
+			/* Registered but no invalidate entry point: MUST
+			 * unregister before failing the load, or the IB
+			 * core keeps vtable pointers into an unloaded
+			 * module (use-after-free on its next peer probe). */
+			ib_unregister_peer_memory_client(ib_reg_handle);
+			ib_reg_handle = NULL;
+		}
 		rp_err("peer-memory registration failed — OFED peer_mem ABI drift?\n");
 		return -EINVAL;
 	}

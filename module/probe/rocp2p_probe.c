@@ -73,6 +73,14 @@ MODULE_VERSION("2.0");
 
 static const struct amd_rdma_interface *rdma_interface;
 
+/* Device-node mode.  Default 0600 (root only): the probe lets its user
+ * pin unbounded GPU memory and mmap BAR pages — a diagnostic surface,
+ * not a service.  Deployments that want group access relax it
+ * explicitly (insmod rocp2p_probe.ko devmode=0660 + a udev group rule);
+ * see docs/RUNBOOK.md.  (was 0666.) */
+static ushort devmode = 0600;
+module_param(devmode, ushort, 0444);
+
 struct probe_ctx {
 	struct list_head pins;
 	struct mutex lock;
@@ -409,6 +417,14 @@ static int probe_mmap(struct file *filp, struct vm_area_struct *vma)
 			if (ov_start < ov_end) {
 				u64 pa = sg_dma_address(sg) +
 					 (ov_start - seg_start);
+				if (pa & ~PAGE_MASK) {
+					/* Sub-page offset would be silently
+					 * truncated by the pfn shift and map
+					 * the wrong bytes: refuse (pin the
+					 * range page-aligned instead). */
+					ret = -EINVAL;
+					break;
+				}
 				ret = io_remap_pfn_range(
 					vma, vaddr,
 					(unsigned long)(pa >> PAGE_SHIFT),
@@ -440,7 +456,7 @@ static struct miscdevice probe_dev = {
 	.minor = MISC_DYNAMIC_MINOR,
 	.name = ROCP2P_PROBE_DEVICE_NAME,
 	.fops = &probe_fops,
-	.mode = 0666,
+	.mode = 0600,	/* overridden by the devmode param at init */
 };
 
 static int __init probe_init(void)
@@ -457,6 +473,7 @@ static int __init probe_init(void)
 		rdma_interface->get_pages, rdma_interface->put_pages,
 		rdma_interface->is_gpu_address,
 		rdma_interface->get_page_size);
+	probe_dev.mode = devmode;
 	ret = misc_register(&probe_dev);
 	if (ret)
 		pp_err("misc_register failed: %d\n", ret);

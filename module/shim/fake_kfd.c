@@ -33,6 +33,12 @@ static uint64_t fk_next_va = 0x700000000000ULL;
 static uint64_t fk_next_bus = 0xd000000000ULL;
 static long fk_stat_get, fk_stat_put, fk_stat_bad_put, fk_stat_cb;
 static int fk_fail_page_size;
+static int fk_reject_null_dev;
+static int fk_revoke_in_get_pages;
+static unsigned int fk_bus_skew;
+
+static void fk_destroy_pin_locked(struct fk_pin *p);
+static void fk_tomb_add_locked(const void *info);
 
 uint64_t fake_kfd_alloc(uint64_t size, unsigned int frag_every)
 {
@@ -43,7 +49,7 @@ uint64_t fake_kfd_alloc(uint64_t size, unsigned int frag_every)
 	a->va = fk_next_va;
 	a->size = size;
 	a->frag_every = frag_every;
-	a->bus_base = fk_next_bus;
+	a->bus_base = fk_next_bus + fk_bus_skew;
 	nchunks = (size + FAKE_KFD_VRAM_PAGE - 1) / FAKE_KFD_VRAM_PAGE;
 	fk_next_va += (size + (1ULL << 30)) & ~((1ULL << 21) - 1);
 	/* leave a hole after the allocation in bus space too */
@@ -98,11 +104,14 @@ static int fk_get_pages(uint64_t address, uint64_t length, struct pid *pid,
 	uint64_t first, last, nchunks, i, off, end;
 	struct scatterlist *sg;
 
-#if ROCNR_AMD_RDMA_HAS_DMA_DEV
-	(void)dma_dev;
-#endif
 	(void)pid;
 	pthread_mutex_lock(&fk_lock);
+#if ROCNR_AMD_RDMA_HAS_DMA_DEV
+	if (fk_reject_null_dev && !dma_dev) {
+		pthread_mutex_unlock(&fk_lock);
+		return -EINVAL;	/* drift: KFD requires the DMA device */
+	}
+#endif
 	a = fk_find_locked(address);
 	if (!a || address + length > a->va + a->size || !length) {
 		pthread_mutex_unlock(&fk_lock);
@@ -129,6 +138,12 @@ static int fk_get_pages(uint64_t address, uint64_t length, struct pid *pid,
 			end = address - a->va + length;
 		sg->dma_address = fk_chunk_bus(a, first + i) +
 				  (off - (first + i) * FAKE_KFD_VRAM_PAGE);
+#if ROCNR_AMD_RDMA_HAS_DMA_DEV
+		/* Drift mode with a real device: KFD device-maps the pin
+		 * internally (the addresses are iovas for dma_dev). */
+		if (fk_reject_null_dev && dma_dev)
+			sg->dma_address += dma_dev->iova_offset;
+#endif
 		sg->dma_length = (unsigned int)(end - off);
 		sg->length = sg->dma_length;
 	}
@@ -139,6 +154,27 @@ static int fk_get_pages(uint64_t address, uint64_t length, struct pid *pid,
 	fk_pins = p;
 	fk_stat_get++;
 	*amd_p2p_data = &p->info;
+
+	if (fk_revoke_in_get_pages > 0) {
+		/* Buffer freed while the pin was being installed: fire the
+		 * free callback synchronously BEFORE get_pages returns, then
+		 * reclaim — the caller gets a pointer to a dead pin and must
+		 * not touch it (its own state machine has seen the revoke). */
+		fk_revoke_in_get_pages--;
+		p->dying = 1;
+		pthread_mutex_unlock(&fk_lock);
+		if (free_callback) {
+			fk_stat_cb++;
+			free_callback(client_priv);
+		}
+		pthread_mutex_lock(&fk_lock);
+		fk_tomb_add_locked(&p->info);
+		fk_destroy_pin_locked(p);
+		pthread_cond_broadcast(&fk_cb_done);
+		pthread_mutex_unlock(&fk_lock);
+		return 0;
+	}
+
 	pthread_mutex_unlock(&fk_lock);
 	return 0;
 }
@@ -258,6 +294,27 @@ void fake_kfd_fail_page_size(int n)
 {
 	pthread_mutex_lock(&fk_lock);
 	fk_fail_page_size = n;
+	pthread_mutex_unlock(&fk_lock);
+}
+
+void fake_kfd_bus_skew(unsigned int bytes)
+{
+	pthread_mutex_lock(&fk_lock);
+	fk_bus_skew = bytes;
+	pthread_mutex_unlock(&fk_lock);
+}
+
+void fake_kfd_reject_null_dev(int on)
+{
+	pthread_mutex_lock(&fk_lock);
+	fk_reject_null_dev = on;
+	pthread_mutex_unlock(&fk_lock);
+}
+
+void fake_kfd_revoke_in_get_pages(int n)
+{
+	pthread_mutex_lock(&fk_lock);
+	fk_revoke_in_get_pages = n;
 	pthread_mutex_unlock(&fk_lock);
 }
 

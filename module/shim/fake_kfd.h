@@ -50,6 +50,28 @@ long fake_kfd_callbacks_fired(void);
  * bridge's documented 2 MiB fallback + warning path). */
 void fake_kfd_fail_page_size(int n);
 
+/* Skew the bus base of subsequent fake allocations by `bytes` (may be a
+ * non-page multiple): models BAR placements whose sub-page alignment
+ * differs from the GPU VA's, which the probe's mmap must refuse to map
+ * rather than truncate.  0 restores natural (2 MiB) alignment. */
+void fake_kfd_bus_skew(unsigned int bytes);
+
+/* Drift simulation: when on, get_pages with dma_dev==NULL
+ * fails with -EINVAL — modeling a ROCK KFD that dma-maps against the
+ * passed device inside get_pages and refuses a NULL one.  When a device
+ * IS passed, the sg table's addresses get that device's iova_offset
+ * applied (the "KFD mapped it internally" shape).  Only meaningful with
+ * ROCNR_AMD_RDMA_HAS_DMA_DEV=1 (the default shim build). */
+void fake_kfd_reject_null_dev(int on);
+
+/* Fault injection: the next N successful get_pages calls fire the pin's
+ * free_callback SYNCHRONOUSLY before returning (buffer freed while the
+ * pin was being installed), then reclaim the pin — the returned
+ * amd_p2p_info is already dead and must not be dereferenced.  Tests the
+ * bridge's lock-free PINNING install (a callback that re-takes the
+ * registration lock from inside get_pages must not deadlock). */
+void fake_kfd_revoke_in_get_pages(int n);
+
 #ifdef __cplusplus
 }
 #endif

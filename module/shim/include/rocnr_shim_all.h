@@ -40,6 +40,7 @@ extern "C" {
 typedef uint8_t  u8;
 typedef uint16_t u16;
 typedef uint32_t u32;
+typedef unsigned short ushort;
 /* unsigned long long (not uint64_t) so %llx format strings match the
  * kernel's u64 without warnings on LP64 userspace */
 typedef unsigned long long u64;

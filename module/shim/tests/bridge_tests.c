@@ -364,6 +364,73 @@ static void test_page_size_fallback(void)
 	printf("ok: page-size query failure falls back to 2 MiB\n");
 }
 
+static void test_null_dev_fallback(void)
+{
+	/* a modern ROCK KFD may refuse get_pages with
+	 * dma_dev=NULL.  The bridge must fall back to deferring the pin to
+	 * dma_map (where the HCA device is known), use KFD's device-mapped
+	 * table verbatim (no dma_map_resource), and still handle dereg and
+	 * invalidation. */
+	uint64_t va, va2;
+	struct fake_ib_mr *mr = NULL;
+	struct sg_table sgt2 = { 0 };
+	int nmap2 = 0;
+	long inv0;
+
+	fake_kfd_reject_null_dev(1);
+
+	/* lifecycle via dereg */
+	va = fake_kfd_alloc(32 * MiB, 0);
+	CHECK(fake_ib_reg_mr(va, 32 * MiB, &dev1, &mr) == 0);
+	CHECK(mr->nmap == 1);
+	/* KFD applied the device mapping internally */
+	CHECK(sg_dma_address(mr->sgt.sgl) >= dev1.iova_offset);
+	/* and the bridge did NOT dma_map_resource on top */
+	CHECK(atomic64_read(&dev1.live_maps) == 0);
+	/* a second HCA shares the table (reference-style, warns) */
+	CHECK(fake_ib_mr_map_also(mr, &dev2, &sgt2, &nmap2) == 0);
+	CHECK(nmap2 == 1);
+	CHECK(sg_dma_address(sgt2.sgl) == sg_dma_address(mr->sgt.sgl));
+	CHECK(fake_ib_mr_unmap_also(mr, &dev2, &sgt2) == 0);
+	CHECK(fake_ib_dereg_mr(mr) == 0);
+	free(mr);
+	fake_kfd_free(va);
+
+	/* lifecycle via invalidation */
+	va2 = fake_kfd_alloc(8 * MiB, 0);
+	mr = NULL;
+	inv0 = fake_ib_invalidate_count();
+	CHECK(fake_ib_reg_mr(va2, 8 * MiB, &dev1, &mr) == 0);
+	fake_kfd_free(va2);
+	CHECK(fake_ib_invalidate_count() == inv0 + 1);
+	CHECK(fake_ib_dereg_mr(mr) == 0);
+	free(mr);
+
+	fake_kfd_reject_null_dev(0);
+	check_balances(devs, 2);
+	printf("ok: dma_dev=NULL drift fallback (deferred pin at dma_map)\n");
+}
+
+static void test_revoke_during_pin(void)
+{
+	/* KFD fires the free callback synchronously from
+	 * inside get_pages.  With the lock-free PINNING install this must
+	 * neither deadlock nor touch the reclaimed pin; the registration
+	 * fails cleanly. */
+	uint64_t va = fake_kfd_alloc(8 * MiB, 0);
+	struct fake_ib_mr *mr = NULL;
+	long cb0 = fake_kfd_callbacks_fired();
+
+	fake_kfd_revoke_in_get_pages(1);
+	CHECK(fake_ib_reg_mr(va, 8 * MiB, &dev1, &mr) == -ENODEV);
+	CHECK(fake_kfd_callbacks_fired() == cb0 + 1);
+	CHECK(fake_kfd_bad_put_calls() == 0);
+	fake_kfd_revoke_in_get_pages(0);
+	fake_kfd_free(va);
+	check_balances(devs, 2);
+	printf("ok: synchronous revoke inside get_pages (no deadlock, clean fail)\n");
+}
+
 static void test_huge_pin(void)
 {
 	/* 64 GiB pin (288 GB HBM sizing): 32768 chunks, hole every 1024 */
@@ -421,6 +488,8 @@ int main(void)
 	test_invalidate();
 	test_invalidate_race();
 	test_page_size_fallback();
+	test_null_dev_fallback();
+	test_revoke_during_pin();
 	test_huge_pin();
 	test_bar_aperture_probe();
 

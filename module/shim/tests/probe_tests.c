@@ -273,6 +273,35 @@ static void test_mmap_maps_every_segment(void)
 	printf("ok: mmap maps all segments, honors sub-ranges, rejects overrun\n");
 }
 
+static void test_mmap_rejects_subpage_offset(void)
+{
+	/* a pin whose BAR address lands at a sub-page offset
+	 * must be refused (-EINVAL), not silently truncated by the pfn
+	 * shift (which would map the wrong bytes).  Skew the fake BAR by
+	 * 0x100 so every sg_dma_address is sub-page misaligned. */
+	struct file *f = dev_open();
+	uint64_t va;
+	struct rocp2p_probe_pin pin;
+	struct vm_area_struct vma;
+
+	fake_kfd_bus_skew(0x100);
+	va = fake_kfd_alloc(8 * MiB, 0);
+	pin.addr = va;
+	pin.length = 4 * MiB;
+	CHECK(xioctl(f, ROCP2P_PROBE_GET_PAGES, &pin) == 0);
+	rocnr_shim_maps_reset();
+	memset(&vma, 0, sizeof(vma));
+	vma.vm_start = 0x70000000;
+	vma.vm_end = 0x70000000 + 1 * MiB;
+	vma.vm_pgoff = va >> PAGE_SHIFT;
+	CHECK(fops->mmap(f, &vma) == -EINVAL);
+	CHECK(rocnr_shim_maps_count() == 0);
+	dev_close(f);
+	fake_kfd_free(va);
+	fake_kfd_bus_skew(0);
+	printf("ok: mmap rejects sub-page BAR offsets (-EINVAL, no truncation)\n");
+}
+
 int main(void)
 {
 	CHECK(rocnr_shim_module_init() == 0);
@@ -287,6 +316,7 @@ int main(void)
 	test_revoke_unlinks();
 	test_revoke_vs_put_race();
 	test_mmap_maps_every_segment();
+	test_mmap_rejects_subpage_offset();
 
 	rocnr_shim_module_exit();
 	CHECK(rocnr_shim_misc_dev() == NULL);

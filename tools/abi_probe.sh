@@ -61,6 +61,11 @@ else
     if grep -E 'get_pages' -A6 "$AMDR" | grep -qE 'struct[[:space:]]+device[[:space:]]*\*'; then
         HAS_DMADEV=1
         echo "  get_pages takes struct device* (modern ROCK)"
+        echo "  NOTE: whether this KFD accepts dma_dev=NULL (bridge maps"
+        echo "  per-HCA later) is NOT probeable from the header.  If pins"
+        echo "  fail at registration with the bridge loaded, the bridge's"
+        echo "  null_dev_fallback (default on) defers the pin to dma_map"
+        echo "  and logs 'amd_rdma ABI drift fallback' — check dmesg."
     else
         HAS_DMADEV=0
         echo "  get_pages has NO dma_dev parameter (legacy KFD)"
