@@ -28,16 +28,20 @@
 #ifndef ROCNR_AMD_RDMA_H_
 #define ROCNR_AMD_RDMA_H_
 
+#ifndef ROCNR_AMD_RDMA_HAS_DMA_DEV
+#define ROCNR_AMD_RDMA_HAS_DMA_DEV 1
+#endif
+
 #ifdef ROCNR_USE_SYSTEM_AMD_RDMA
+/* Real ROCK tree (or the third_party/ reconstructions).  The
+ * ROCNR_AMD_RDMA_HAS_DMA_DEV switch must be set to match the tree's
+ * get_pages signature; the bridge's call site compiles against it, so
+ * a wrong setting is a hard compile error (arg-count mismatch). */
 #include <drm/amd_rdma.h>
 #else
 
 #include <linux/types.h>
 #include <linux/scatterlist.h>
-
-#ifndef ROCNR_AMD_RDMA_HAS_DMA_DEV
-#define ROCNR_AMD_RDMA_HAS_DMA_DEV 1
-#endif
 
 struct pid;
 struct device;

@@ -28,8 +28,22 @@
 #ifndef ROCNR_PEER_MEM_H_
 #define ROCNR_PEER_MEM_H_
 
+#ifndef ROCNR_PEER_MEM_CORE_CONTEXT_U64
+#define ROCNR_PEER_MEM_CORE_CONTEXT_U64 1
+#endif
+
 #ifdef ROCNR_USE_SYSTEM_PEER_MEM
+/* Real OFED tree (or the third_party/ reconstructions): take every
+ * declaration from it; only the registration-cookie typedef the bridge
+ * uses in its own signatures is added on top (it must agree with the
+ * system header's get_pages — the compile matrix builds with
+ * -Werror=incompatible-pointer-types so a mismatch is a hard error). */
 #include <rdma/peer_mem.h>
+#if ROCNR_PEER_MEM_CORE_CONTEXT_U64
+typedef u64 rocnr_core_context_t;
+#else
+typedef void *rocnr_core_context_t;
+#endif
 #else
 
 #include <linux/types.h>
@@ -37,10 +51,6 @@
 
 #define IB_PEER_MEMORY_NAME_MAX  64
 #define IB_PEER_MEMORY_VER_MAX   16
-
-#ifndef ROCNR_PEER_MEM_CORE_CONTEXT_U64
-#define ROCNR_PEER_MEM_CORE_CONTEXT_U64 1
-#endif
 
 #if ROCNR_PEER_MEM_CORE_CONTEXT_U64
 typedef u64 rocnr_core_context_t;
@@ -92,6 +102,30 @@ struct peer_memory_client {
 	void (*put_pages)(struct sg_table *sg_head, void *client_context);
 	unsigned long (*get_page_size)(void *client_context);
 	void (*release)(void *client_context);
+	/* Optional per-peer-id private-data hooks — present in the real
+	 * MOFED struct (third_party/mlnx_ofed/rdma/peer_mem.h); a client
+	 * that omitted them from a vendored layout would hand the core a
+	 * TOO-SHORT struct (round-1 contract-fidelity gap, now closed).
+	 * May be NULL. */
+	void* (*get_context_private_data)(u64 peer_id);
+	void (*put_context_private_data)(void *context);
+};
+
+enum {
+	PEER_MEM_INVALIDATE_UNMAPS = 1 << 0,
+};
+
+/* Extended registration (nvidia-peermem / MOFED 5.x generation): the
+ * core recognizes a peer_memory_client embedded at the head of a
+ * peer_memory_client_ex by ex_size and reads capability flags from it.
+ * PEER_MEM_INVALIDATE_UNMAPS: this client's invalidation flow already
+ * unmaps (our free callback revokes the pin and the dma_unmap that the
+ * core's teardown then performs is a no-op re-entry), letting the core
+ * skip redundant unmap work. */
+struct peer_memory_client_ex {
+	struct peer_memory_client client;
+	size_t ex_size;
+	u32 flags;
 };
 
 void *ib_register_peer_memory_client(const struct peer_memory_client *client,

@@ -364,6 +364,7 @@ static void test_page_size_fallback(void)
 	printf("ok: page-size query failure falls back to 2 MiB\n");
 }
 
+#if ROCNR_AMD_RDMA_HAS_DMA_DEV
 static void test_null_dev_fallback(void)
 {
 	/* a modern ROCK KFD may refuse get_pages with
@@ -410,6 +411,12 @@ static void test_null_dev_fallback(void)
 	check_balances(devs, 2);
 	printf("ok: dma_dev=NULL drift fallback (deferred pin at dma_map)\n");
 }
+#else
+static void test_null_dev_fallback(void)
+{
+	printf("ok: dma_dev=NULL drift fallback (n/a: legacy amd_rdma ABI)\n");
+}
+#endif
 
 static void test_revoke_during_pin(void)
 {

@@ -81,3 +81,19 @@ def test_shim_suites_tsan(binary):
                          capture_output=True, text=True, timeout=600)
     assert out.returncode == 0, out.stdout + out.stderr
     assert "WARNING: ThreadSanitizer" not in out.stdout + out.stderr
+
+
+@pytest.mark.skipif(shutil.which("gcc") is None or
+                    shutil.which("make") is None, reason="no toolchain")
+@pytest.mark.timeout(900)
+def test_abi_compile_matrix():
+    """the bridge must compile (and its full suite run)
+    against every generation of the two external contracts — the
+    vendored headers under both drift-switch settings AND the
+    third_party/ reconstructions of the real public MLNX_OFED
+    peer_mem.h / ROCK amd_rdma.h, with signature drift a hard compile
+    error (-Werror=incompatible-pointer-types)."""
+    out = subprocess.run(["make", "-C", SHIM, "matrix"],
+                         capture_output=True, text=True, timeout=800)
+    assert out.returncode == 0, out.stdout + out.stderr
+    assert "ABI MATRIX PASSED" in out.stdout
