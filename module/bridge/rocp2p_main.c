@@ -144,8 +144,20 @@ enum rocp2p_state {
  * (amdp2p.c:222-240).  Additional HCAs mapping the same MR reuse that
  * table with a loud warning (correct only when the IOMMU does not
  * isolate the two HCAs differently).  See docs/LIMITATIONS.md. */
+#if ROCNR_AMD_RDMA_HAS_DMA_DEV
 static bool null_dev_fallback = true;
 module_param(null_dev_fallback, bool, 0444);
+#endif
+
+/* Register with the modern extended surface (peer_memory_client_ex +
+ * PEER_MEM_INVALIDATE_UNMAPS) by default; peer_ex=0 registers the
+ * plain 2016-style client for cores that predate the extension.  The
+ * flag is truthful: our free callback revokes the KFD pin before the
+ * invalidate upcall, so the core's post-invalidate dma_unmap is
+ * redundant (per-HCA IOVA mappings are reclaimed in release()).
+ * Non-static so the userspace shim suite can drive both generations. */
+bool rocp2p_peer_ex = true;
+module_param_named(peer_ex, rocp2p_peer_ex, bool, 0444);
 
 /* One dma_map() result for one device. */
 struct rocp2p_dmamap {
@@ -656,6 +668,25 @@ static void rocp2p_release(void *client_context)
 }
 
 /* ------------------------------------------------------------------ */
+/* Registered as the head of a peer_memory_client_ex so an ex-aware IB
+ * core (MOFED 5.x / nvidia-peermem generation) can read capability
+ * flags; the marker is the public convention of that generation (last
+ * byte of version[] set to 1 — see module/include/rocnr_peer_mem.h).
+ * Older cores read only the embedded plain client. */
+#if ROCNR_PEER_MEM_HAS_EX
+static struct peer_memory_client_ex rocp2p_client_ex = {
+	.client = {
+		.acquire = rocp2p_acquire,
+		.get_pages = rocp2p_get_pages,
+		.dma_map = rocp2p_dma_map,
+		.dma_unmap = rocp2p_dma_unmap,
+		.put_pages = rocp2p_put_pages,
+		.get_page_size = rocp2p_get_page_size,
+		.release = rocp2p_release,
+	},
+};
+#define rocp2p_client rocp2p_client_ex.client
+#else	/* 2016-era tree: plain client only */
 static struct peer_memory_client rocp2p_client = {
 	.acquire = rocp2p_acquire,
 	.get_pages = rocp2p_get_pages,
@@ -665,6 +696,7 @@ static struct peer_memory_client rocp2p_client = {
 	.get_page_size = rocp2p_get_page_size,
 	.release = rocp2p_release,
 };
+#endif
 
 static int __init rocp2p_init(void)
 {
@@ -687,6 +719,21 @@ static int __init rocp2p_init(void)
 		sizeof(rocp2p_client.name));
 	strscpy(rocp2p_client.version, ROCP2P_DRIVER_VERSION,
 		sizeof(rocp2p_client.version));
+#if ROCNR_PEER_MEM_HAS_EX
+	if (rocp2p_peer_ex) {
+		/* ex marker convention: last version byte = 1 */
+		rocp2p_client.version[IB_PEER_MEMORY_VER_MAX - 1] = 1;
+		rocp2p_client_ex.ex_size = sizeof(rocp2p_client_ex);
+		rocp2p_client_ex.flags = PEER_MEM_INVALIDATE_UNMAPS;
+	} else {
+		rocp2p_client.version[IB_PEER_MEMORY_VER_MAX - 1] = 0;
+		rocp2p_client_ex.ex_size = 0;
+		rocp2p_client_ex.flags = 0;
+	}
+#else
+	if (rocp2p_peer_ex)
+		rp_info("peer_ex requested but this peer_mem ABI generation has no extended surface; registering plain client\n");
+#endif
 
 	rocp2p_check_bar_aperture();
 

@@ -32,6 +32,14 @@
 #define ROCNR_PEER_MEM_CORE_CONTEXT_U64 1
 #endif
 
+/* Does the peer_mem ABI generation in use declare the extended
+ * registration surface (peer_memory_client_ex + capability flags)?
+ * 1 for MOFED 5.x / nvidia-peermem-era trees (and the vendored
+ * header); set 0 when building against a 2016-era tree. */
+#ifndef ROCNR_PEER_MEM_HAS_EX
+#define ROCNR_PEER_MEM_HAS_EX 1
+#endif
+
 #ifdef ROCNR_USE_SYSTEM_PEER_MEM
 /* Real OFED tree (or the third_party/ reconstructions): take every
  * declaration from it; only the registration-cookie typedef the bridge

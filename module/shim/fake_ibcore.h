@@ -50,6 +50,19 @@ void fake_ib_reset_stats(void);
 /* The registered client (NULL when none). */
 const struct peer_memory_client *fake_ib_client(void);
 
+/* Modern-core (MOFED 5.x generation) surface: extended-registration
+ * detection and capability flags, readable by tests. */
+int fake_ib_client_is_ex(void);
+u32 fake_ib_client_flags(void);
+
+/* Core-owned invalidation mode: the invalidate entry returns
+ * immediately and the MR teardown (dma_unmap -> put_pages -> release)
+ * runs later on a core-owned thread — the ordering of the newer
+ * rdma-core peer-mem flow.  fake_ib_quiesce() waits for all deferred
+ * teardowns (call before freeing the fake_ib_mr). */
+void fake_ib_set_async_invalidate(int on);
+void fake_ib_quiesce(void);
+
 #ifdef __cplusplus
 }
 #endif

@@ -158,7 +158,7 @@ struct pci_dev *pci_get_device(unsigned int vendor, unsigned int device,
 		i = (int)(from - shim_pci_devs) + 1;
 	for (; i < shim_pci_n; i++) {
 		if (shim_pci_devs[i].vendor == vendor &&
-		    (device == PCI_ANY_ID ||
+		    (device == (unsigned int)PCI_ANY_ID ||
 		     shim_pci_devs[i].device == device))
 			return &shim_pci_devs[i];
 	}

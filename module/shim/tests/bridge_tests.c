@@ -454,6 +454,98 @@ static void test_huge_pin(void)
 	printf("ok: 64 GiB pin -> 32 sg entries\n");
 }
 
+extern bool rocp2p_peer_ex;
+
+static void test_ex_registration(void)
+{
+	/* Modern surface: default registration is a
+	 * peer_memory_client_ex with PEER_MEM_INVALIDATE_UNMAPS; an
+	 * ex-aware core then SKIPS dma_unmap on the invalidate path and
+	 * the bridge's release() must reclaim the per-device mappings. */
+	uint64_t va;
+	struct fake_ib_mr *mr = NULL;
+
+#if ROCNR_PEER_MEM_HAS_EX
+	CHECK(fake_ib_client_is_ex() == 1);
+	CHECK(fake_ib_client_flags() == PEER_MEM_INVALIDATE_UNMAPS);
+#else
+	CHECK(fake_ib_client_is_ex() == 0);
+#endif
+	va = fake_kfd_alloc(16 * MiB, 0);
+	CHECK(fake_ib_reg_mr(va, 16 * MiB, &dev1, &mr) == 0);
+	fake_kfd_free(va);	/* invalidate; ex core skips dma_unmap */
+	CHECK(mr->invalidated == 1);
+	CHECK(atomic64_read(&dev1.live_maps) == 0);	/* release cleaned */
+	CHECK(fake_ib_dereg_mr(mr) == 0);
+	free(mr);
+	check_balances(devs, 2);
+	printf("ok: ex registration + INVALIDATE_UNMAPS (core skips dma_unmap)\n");
+}
+
+static void test_plain_generation(void)
+{
+	/* peer_ex=0: the bridge registers the 2016-style plain client;
+	 * the core sees no ex marker and uses the full teardown. */
+	uint64_t va;
+	struct fake_ib_mr *mr = NULL;
+
+	rocnr_shim_module_exit();
+	rocp2p_peer_ex = false;
+	CHECK(rocnr_shim_module_init() == 0);
+	CHECK(fake_ib_client_is_ex() == 0);
+	CHECK(fake_ib_client_flags() == 0);
+	va = fake_kfd_alloc(8 * MiB, 0);
+	CHECK(fake_ib_reg_mr(va, 8 * MiB, &dev1, &mr) == 0);
+	fake_kfd_free(va);
+	CHECK(mr->invalidated == 1);
+	CHECK(fake_ib_dereg_mr(mr) == 0);
+	free(mr);
+	check_balances(devs, 2);
+	rocnr_shim_module_exit();
+	rocp2p_peer_ex = true;
+	CHECK(rocnr_shim_module_init() == 0);
+	printf("ok: plain-generation registration (peer_ex=0)\n");
+}
+
+static void test_core_owned_invalidation_race(void)
+{
+	/* Newer rdma-core ordering: invalidate() returns immediately and
+	 * the core tears the MR down LATER from its own thread — so KFD
+	 * reclaims the pin long before put_pages/release arrive, while a
+	 * concurrent ibv_dereg_mr may race the deferred teardown. */
+	int ITERS = 1000;
+	const char *env_iters = getenv("ROCNR_RACE_ITERS");
+	unsigned int seed = 777;
+	int i;
+
+	if (env_iters && atoi(env_iters) > 0)
+		ITERS = atoi(env_iters) / 3 + 1;
+	fake_ib_set_async_invalidate(1);
+	for (i = 0; i < ITERS; i++) {
+		struct fake_ib_mr *mr = NULL;
+		struct race_arg a;
+		pthread_t th;
+		struct timespec ts = { 0, 0 };
+
+		a.va = fake_kfd_alloc(4 * MiB, 0);
+		CHECK(fake_ib_reg_mr(a.va, 4 * MiB, &dev1, &mr) == 0);
+		a.delay_ns = rand_r(&seed) % 20000;
+		pthread_create(&th, NULL, race_free_thread, &a);
+		ts.tv_nsec = rand_r(&seed) % 20000;
+		nanosleep(&ts, NULL);
+		fake_ib_dereg_mr(mr);
+		pthread_join(th, NULL);
+		fake_ib_quiesce();
+		fake_ib_dereg_mr(mr);	/* idempotent */
+		free(mr);
+		CHECK(fake_kfd_bad_put_calls() == 0);
+	}
+	fake_ib_set_async_invalidate(0);
+	check_balances(devs, 2);
+	printf("ok: %d core-owned (deferred) invalidation races, clean\n",
+	       ITERS);
+}
+
 static void test_bar_aperture_probe(void)
 {
 	/* init already probed an empty PCI table (warned).  Re-run module
@@ -497,6 +589,9 @@ int main(void)
 	test_page_size_fallback();
 	test_null_dev_fallback();
 	test_revoke_during_pin();
+	test_ex_registration();
+	test_plain_generation();
+	test_core_owned_invalidation_race();
 	test_huge_pin();
 	test_bar_aperture_probe();
 
