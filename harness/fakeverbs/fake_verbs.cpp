@@ -11,6 +11,9 @@
 //                            (default ib) — covers both AH paths.
 #include "infiniband/verbs.h"
 
+#include <sys/mman.h>
+#include <unistd.h>
+
 #include <cstdlib>
 #include <cstring>
 #include <cstdio>
@@ -25,6 +28,12 @@ struct FakeMr {
   ibv_mr mr;
   int access;
   bool live;
+  // dmabuf MRs: CPU window mapped from the fd (real BAR pages when the
+  // fd came from hipMemGetHandleForAddressRange on VRAM; a memfd in
+  // CPU-only CI).  mr.addr holds the iova; data-plane addresses are
+  // translated iova -> host_map.
+  uint8_t* host_map = nullptr;
+  size_t map_len = 0;
 };
 
 struct FakeCq {
@@ -65,6 +74,13 @@ FakeMr* find_mr_by_key(uint32_t key, bool remote) {
 bool range_ok(const FakeMr* m, uint64_t addr, uint64_t len) {
   uint64_t base = (uint64_t)m->mr.addr;
   return addr >= base && addr + len <= base + m->mr.length;
+}
+
+// Data-plane pointer for an in-MR address: identity for host MRs,
+// iova->CPU-window translation for dmabuf MRs.
+uint8_t* mr_data_ptr(const FakeMr* m, uint64_t addr) {
+  if (m->host_map) return m->host_map + (addr - (uint64_t)m->mr.addr);
+  return (uint8_t*)addr;
 }
 
 }  // namespace
@@ -109,19 +125,30 @@ struct ibv_mr* ibv_reg_mr(struct ibv_pd* pd, void* addr, size_t length,
 struct ibv_mr* ibv_reg_dmabuf_mr(struct ibv_pd* pd, uint64_t offset,
                                  size_t length, uint64_t iova, int fd,
                                  int access) {
-  // data plane for dmabuf regions is device memory — the fake records
-  // the registration (iova-addressed) but cannot memcpy it; posts
-  // against it fail loudly so tests keep to host MRs.
+  // Map the dmabuf into this process the way a real HCA would DMA it:
+  // through the exporter's backing pages.  For a VRAM buffer exported
+  // by hipMemGetHandleForAddressRange this is the PCIe BAR window
+  // (amdgpu implements dma-buf mmap), so posts against the MR move
+  // real bytes over the bus; in CPU-only CI the fd is a memfd.
   if (fd < 0) return nullptr;
-  (void)offset;
+  void* map = mmap(nullptr, length + offset, PROT_READ | PROT_WRITE,
+                   MAP_SHARED, fd, 0);
+  if (map == MAP_FAILED) {
+    fprintf(stderr,
+            "fake_verbs: dmabuf fd %d not mmappable (%m) — registration "
+            "refused\n", fd);
+    return nullptr;
+  }
   auto* m = new FakeMr();
   m->mr.pd = pd;
   m->mr.addr = (void*)iova;
   m->mr.length = length;
   m->mr.lkey = g_ctx.next_key++;
   m->mr.rkey = g_ctx.next_key++;
-  m->access = access | (1 << 30);  // mark: not host-addressable
+  m->access = access;
   m->live = true;
+  m->host_map = (uint8_t*)map + offset;
+  m->map_len = length + offset;
   g_ctx.mrs.push_back(m);
   return &m->mr;
 }
@@ -129,6 +156,9 @@ struct ibv_mr* ibv_reg_dmabuf_mr(struct ibv_pd* pd, uint64_t offset,
 int ibv_dereg_mr(struct ibv_mr* mr) {
   for (auto* m : g_ctx.mrs)
     if (&m->mr == mr && m->live) {
+      if (m->host_map)
+        munmap(m->host_map - (m->map_len - m->mr.length), m->map_len);
+      m->host_map = nullptr;
       m->live = false;
       return 0;
     }
@@ -254,20 +284,18 @@ int ibv_post_send(struct ibv_qp* qp, struct ibv_send_wr* wr,
     if (!remote) FAIL("post_send: bad rkey");
     if (!range_ok(remote, wr->wr.rdma.remote_addr, wr->sg_list[0].length))
       FAIL("post_send: remote range out of MR bounds");
-    if (remote->access & (1 << 30))
-      FAIL("post_send: fake layer cannot address dmabuf (device) MRs");
+    uint8_t* rptr = mr_data_ptr(remote, wr->wr.rdma.remote_addr);
+    uint8_t* lptr = mr_data_ptr(local, wr->sg_list[0].addr);
     if (wr->opcode == IBV_WR_RDMA_WRITE) {
       if (!(remote->access & IBV_ACCESS_REMOTE_WRITE))
         FAIL("post_send: remote MR lacks REMOTE_WRITE");
-      memcpy((void*)wr->wr.rdma.remote_addr, (void*)wr->sg_list[0].addr,
-             wr->sg_list[0].length);
+      memcpy(rptr, lptr, wr->sg_list[0].length);
     } else if (wr->opcode == IBV_WR_RDMA_READ) {
       if (!(remote->access & IBV_ACCESS_REMOTE_READ))
         FAIL("post_send: remote MR lacks REMOTE_READ");
       if (!(local->access & IBV_ACCESS_LOCAL_WRITE))
         FAIL("post_send: local MR lacks LOCAL_WRITE");
-      memcpy((void*)wr->sg_list[0].addr, (void*)wr->wr.rdma.remote_addr,
-             wr->sg_list[0].length);
+      memcpy(lptr, rptr, wr->sg_list[0].length);
     } else {
       FAIL("post_send: unsupported opcode");
     }

@@ -21,6 +21,7 @@ struct TransportConfig {
   int device_index = 0;      // GPU ordinal (hip/verbs-gpu backends)
   int num_streams = 2;       // hip stream engine
   size_t inflight = 0;       // 0 = backend-chosen
+  size_t chain = 0;          // verbs: WRs per doorbell (0 = auto)
   std::string engine = "auto";  // hip: auto|kernel|stream
   std::string verbs_mr = "auto";  // verbs: auto|peer|dmabuf|host
   bool wc_staging = false;   // hip: write-combined pinned staging

@@ -138,6 +138,7 @@ int main(int argc, char** argv) {
     else if (a == "--mr") cfg.verbs_mr = next();
     else if (a == "--streams") cfg.num_streams = atoi(next().c_str());
     else if (a == "--inflight") cfg.inflight = strtoull(next().c_str(), 0, 0);
+    else if (a == "--chain") cfg.chain = strtoull(next().c_str(), 0, 0);
     else if (a == "--seed") seed = strtoull(next().c_str(), 0, 0);
     else if (a == "--wc") cfg.wc_staging = true;
     else if (a == "--serve") serve_port = atoi(next().c_str());

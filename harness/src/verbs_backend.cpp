@@ -66,8 +66,10 @@ bool verbs_runtime_available() {
 #ifdef ROCP2P_HAVE_VERBS
 #include <infiniband/verbs.h>
 #include <hip/hip_runtime.h>
+#include <sys/mman.h>
 #include <unistd.h>
 
+#include <algorithm>
 #include <cstdio>
 #include <vector>
 
@@ -210,6 +212,7 @@ struct Region {
   uint8_t* ptr = nullptr;
   ibv_mr* mr = nullptr;
   size_t bytes = 0;
+  bool host_is_map = false;  // host came from mmap (memfd dmabuf stand-in)
   std::string mode;
 
   void create(ibv_pd* pd, size_t nbytes, std::string mr_mode,
@@ -228,24 +231,44 @@ struct Region {
       VB_THROW(mr, "ibv_reg_mr(region) failed");
       return;
     }
+    if (mode == "dmabuf") {
+      int fd = -1;
+      if (hip_available()) {
+        // VRAM dmabuf: the exporter's pages ARE the HBM region (BAR
+        // window); module-free cross-check of the peer-MR path.
+        VB_THROW(hipSetDevice(device_index) == hipSuccess, "hipSetDevice");
+        VB_THROW(hipMalloc((void**)&gpu, nbytes) == hipSuccess,
+                 "hipMalloc region");
+        ptr = gpu;
+        VB_THROW(hipMemGetHandleForAddressRange(
+                     &fd, gpu, nbytes, hipMemRangeHandleTypeDmaBufFd, 0) ==
+                     hipSuccess,
+                 "hipMemGetHandleForAddressRange(dmabuf) failed");
+      } else {
+        // CPU CI: a memfd plays the exporter so the ibv_reg_dmabuf_mr
+        // code path (fd handoff, iova addressing, MR lifetime) still
+        // executes end-to-end.
+        fd = (int)memfd_create("rocp2p_dmabuf", 0);
+        VB_THROW(fd >= 0, "memfd_create failed");
+        VB_THROW(ftruncate(fd, (off_t)nbytes) == 0, "ftruncate failed");
+        host = (uint8_t*)mmap(nullptr, nbytes, PROT_READ | PROT_WRITE,
+                              MAP_SHARED, fd, 0);
+        VB_THROW(host != MAP_FAILED, "mmap(memfd) failed");
+        host_is_map = true;
+        ptr = host;
+      }
+      mr = ibv_reg_dmabuf_mr(pd, 0, nbytes, (uint64_t)ptr, fd, acc);
+      close(fd);  // the MR holds its own reference to the dmabuf
+      VB_THROW(mr, "ibv_reg_dmabuf_mr failed");
+      return;
+    }
+    // peer mode: dispatched to the rocp2p bridge by the IB core
     VB_THROW(hipSetDevice(device_index) == hipSuccess, "hipSetDevice");
     VB_THROW(hipMalloc((void**)&gpu, nbytes) == hipSuccess,
              "hipMalloc region");
     ptr = gpu;
-    if (mode == "dmabuf") {
-      int fd = -1;
-      VB_THROW(hipMemGetHandleForAddressRange(
-                   &fd, gpu, nbytes, hipMemRangeHandleTypeDmaBufFd, 0) ==
-                   hipSuccess,
-               "hipMemGetHandleForAddressRange(dmabuf) failed");
-      mr = ibv_reg_dmabuf_mr(pd, 0, nbytes, (uint64_t)gpu, fd, acc);
-      close(fd);  // the MR holds its own reference to the dmabuf
-      VB_THROW(mr, "ibv_reg_dmabuf_mr failed");
-    } else {
-      // peer mode: dispatched to the rocp2p bridge by the IB core
-      mr = ibv_reg_mr(pd, ptr, nbytes, acc);
-      VB_THROW(mr, "ibv_reg_mr(GPU VA) failed — rocp2p bridge loaded?");
-    }
+    mr = ibv_reg_mr(pd, ptr, nbytes, acc);
+    VB_THROW(mr, "ibv_reg_mr(GPU VA) failed — rocp2p bridge loaded?");
   }
 
   uint64_t verify(uint64_t seed) {
@@ -281,10 +304,14 @@ struct Region {
 
   void destroy() {
     if (mr) ibv_dereg_mr(mr);
-    if (host) free(host);
+    if (host) {
+      if (host_is_map) munmap(host, bytes);
+      else free(host);
+    }
     if (gpu) (void)hipFree(gpu);
     mr = nullptr;
     host = gpu = ptr = nullptr;
+    host_is_map = false;
   }
 };
 
@@ -306,6 +333,10 @@ class VerbsTransport final : public Transport {
       throw std::runtime_error("region must be a multiple of msg size");
     inflight_ = cfg.inflight ? cfg.inflight : 64;
     if (inflight_ > msgs_per_region()) inflight_ = msgs_per_region();
+    chain_ = cfg.chain ? cfg.chain : std::min<size_t>(16, inflight_);
+    if (chain_ > inflight_) chain_ = inflight_;
+    wrs_.resize(chain_);
+    sges_.resize(chain_);
 
     ctx_ = open_first_device();
     pd_ = ibv_alloc_pd(ctx_);
@@ -383,27 +414,43 @@ class VerbsTransport final : public Transport {
     return oob_ ? "verbs-client" : "verbs";
   }
 
+  // WR chaining + selective signaling (the mlx5 doorbell-batching
+  // shape): one ibv_post_send call rings ONE doorbell for a chain of
+  // up to chain_ WRs, and only the chain's last WR is SIGNALED — its
+  // wr_id carries the chain length so drain() retires the whole chain
+  // per completion.  Posting one SIGNALED WR per call (round-1 shape)
+  // would cap a real HCA's 4 KiB message rate at the per-doorbell +
+  // per-CQE cost; see docs/PERF.md "posting rate".
   void post_many(uint64_t start, uint64_t n) override {
-    for (uint64_t i = start; i < start + n; i++) {
+    uint64_t i = start;
+    while (i < start + n) {
       if (outstanding_ >= inflight_) drain(1);
-      ibv_sge sge;
-      sge.addr = (uint64_t)(staging_ + (i % inflight_) * cfg_.msg_bytes);
-      sge.length = (uint32_t)cfg_.msg_bytes;
-      sge.lkey = staging_mr_->lkey;
-      ibv_send_wr wr;
-      memset(&wr, 0, sizeof(wr));
-      wr.wr_id = i;
-      wr.sg_list = &sge;
-      wr.num_sge = 1;
-      wr.opcode = cfg_.dir == Direction::Write ? IBV_WR_RDMA_WRITE
-                                               : IBV_WR_RDMA_READ;
-      wr.send_flags = IBV_SEND_SIGNALED;
-      wr.wr.rdma.remote_addr =
-          remote_addr_ + (i % msgs_per_region()) * cfg_.msg_bytes;
-      wr.wr.rdma.rkey = remote_rkey_;
+      uint64_t room = (uint64_t)(inflight_ - outstanding_);
+      uint64_t chain = std::min({start + n - i, room, (uint64_t)chain_});
+      for (uint64_t k = 0; k < chain; k++) {
+        uint64_t m = i + k;
+        sges_[k].addr =
+            (uint64_t)(staging_ + (m % inflight_) * cfg_.msg_bytes);
+        sges_[k].length = (uint32_t)cfg_.msg_bytes;
+        sges_[k].lkey = staging_mr_->lkey;
+        memset(&wrs_[k], 0, sizeof(ibv_send_wr));
+        wrs_[k].wr_id = 0;  // unsignaled: retired by the chain tail
+        wrs_[k].next = k + 1 < chain ? &wrs_[k + 1] : nullptr;
+        wrs_[k].sg_list = &sges_[k];
+        wrs_[k].num_sge = 1;
+        wrs_[k].opcode = cfg_.dir == Direction::Write ? IBV_WR_RDMA_WRITE
+                                                      : IBV_WR_RDMA_READ;
+        wrs_[k].send_flags = 0;
+        wrs_[k].wr.rdma.remote_addr =
+            remote_addr_ + (m % msgs_per_region()) * cfg_.msg_bytes;
+        wrs_[k].wr.rdma.rkey = remote_rkey_;
+      }
+      wrs_[chain - 1].send_flags = IBV_SEND_SIGNALED;
+      wrs_[chain - 1].wr_id = chain;
       ibv_send_wr* bad = nullptr;
-      VB_THROW(ibv_post_send(qp_, &wr, &bad) == 0, "ibv_post_send");
-      outstanding_++;
+      VB_THROW(ibv_post_send(qp_, &wrs_[0], &bad) == 0, "ibv_post_send");
+      outstanding_ += chain;
+      i += chain;
     }
   }
 
@@ -453,18 +500,23 @@ class VerbsTransport final : public Transport {
   }
 
  private:
+  // at_least counts WRs; each completion retires wr_id WRs (its chain).
   void drain(size_t at_least) {
     ibv_wc wc[16];
     size_t done = 0;
     while (done < at_least && outstanding_ > 0) {
       int n = ibv_poll_cq(cq_, 16, wc);
       VB_THROW(n >= 0, "ibv_poll_cq");
-      for (int i = 0; i < n; i++)
+      for (int i = 0; i < n; i++) {
         VB_THROW(wc[i].status == IBV_WC_SUCCESS,
                  std::string("completion error: ") +
                      ibv_wc_status_str(wc[i].status));
-      done += (size_t)n;
-      outstanding_ -= (size_t)n;
+        size_t retired = (size_t)wc[i].wr_id;
+        VB_THROW(retired >= 1 && retired <= outstanding_,
+                 "completion retires more WRs than outstanding");
+        done += retired;
+        outstanding_ -= retired;
+      }
     }
   }
 
@@ -480,7 +532,10 @@ class VerbsTransport final : public Transport {
   Region region_;  // loopback only
   uint64_t remote_addr_ = 0;
   uint32_t remote_rkey_ = 0;
-  size_t outstanding_ = 0;
+  size_t outstanding_ = 0;	/* WRs in the SQ (signaled or not) */
+  size_t chain_ = 1;		/* WRs per doorbell */
+  std::vector<ibv_send_wr> wrs_;
+  std::vector<ibv_sge> sges_;
 };
 
 std::unique_ptr<Transport> make_verbs_transport(const TransportConfig& cfg) {
