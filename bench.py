@@ -30,6 +30,27 @@ METRIC = ("ib_write_bw GB/s NIC<->GPU-HBM at 4KB/1MB/64MB msg; "
           "1/2/4/8 MI355X aggregate")
 
 
+def verbs_ok() -> bool:
+    """A usable verbs stack: libibverbs + >=1 IB device + the built
+    native harness (ROCNR_FORCE_VERBS=1 forces it for the fake-verbs
+    CI tier)."""
+    if os.environ.get("ROCNR_FORCE_VERBS") == "1":
+        return True
+    from rocnrdma_amd.transport.verbs import harness_binary, verbs_available
+
+    return verbs_available() and harness_binary() is not None
+
+
+def choose_transport(has_gpu: bool, verbs: bool) -> str:
+    """--transport auto order: verbs — the BASELINE
+    metric's real data plane (ibv_reg_mr + RDMA write, reference
+    README.md:67) — then sdma (same PCIe/BAR path an HCA masters),
+    then fake."""
+    if verbs:
+        return "verbs"
+    return "sdma" if has_gpu else "fake"
+
+
 def parse_args():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
@@ -43,6 +64,10 @@ def parse_args():
                    help="0 = one full region pass per step")
     p.add_argument("--transport", default="auto",
                    choices=["auto", "fake", "sdma", "verbs"])
+    p.add_argument("--verbs-mr", default="auto",
+                   choices=["auto", "peer", "dmabuf", "host"],
+                   help="verbs MR mode (auto: peer on GPU — the bridge "
+                        "path — else host)")
     p.add_argument("--direction", default="write",
                    choices=["write", "read"])
     p.add_argument("--inflight", type=int, default=0,
@@ -88,13 +113,16 @@ def main():
     region = args.region_bytes
     tname = args.transport
     if tname == "auto":
-        # verbs measurement runs through the native harness
-        # (harness/build/rocp2p_bw) on HCA hosts; the python bench
-        # measures the PCIe/BAR layer directly
-        tname = "sdma" if has_gpu else "fake"
+        tname = choose_transport(has_gpu, verbs_ok())
     if tname == "fake" and args.region_bytes > (64 << 20):
         region = min(region, 256 << 20)
+    if tname == "verbs" and not has_gpu and args.region_bytes > (64 << 20):
+        region = min(region, 256 << 20)   # CPU fake-verbs tier
     msg = min(args.msg_bytes, region)
+
+    if tname == "verbs":
+        return run_verbs(args, msg, region, rank, world, has_gpu, device,
+                         numa, dist)
 
     tp = get_transport(tname, msg_bytes=msg, region_bytes=region,
                        inflight=args.inflight, direction=args.direction,
@@ -182,6 +210,16 @@ def main():
     if msg_sweep is not None:
         msg_sweep[str(msg)] = round(value, 3)
 
+    tp.close()
+    emit_result(args, rank, world, has_gpu, value=value, elapsed=elapsed,
+                msgs_per_step=msgs_per_step, msg=msg, region=region,
+                transport=tp.name, integrity=integrity, numa=numa,
+                msg_sweep=msg_sweep, dist=dist)
+
+
+def emit_result(args, rank, world, has_gpu, *, value, elapsed,
+                msgs_per_step, msg, region, transport, integrity, numa,
+                msg_sweep, dist, mr=None):
     result = {
         "metric": METRIC,
         "value": round(value, 3),
@@ -200,7 +238,7 @@ def main():
             "global_batch": msgs_per_step * world,
             "seq_len": msg,
             "parallelism": f"1qp-per-gpu x{world}",
-            "transport": tp.name,
+            "transport": transport,
             "direction": args.direction,
             "msg_bytes": msg,
             "region_bytes": region,
@@ -211,17 +249,76 @@ def main():
             "msg_sweep_gbps": msg_sweep,
         },
     }
+    if mr is not None:
+        result["config"]["verbs_mr"] = mr
     if rank == 0:
         line = json.dumps(result)
         print(line)
         if args.json_out:
             with open(args.json_out, "w") as f:
                 f.write(line + "\n")
-    tp.close()
     if dist is not None:
         dist.destroy_process_group()
     if integrity.startswith("FAILED"):
         raise SystemExit(2)
+
+
+def run_verbs(args, msg, region, rank, world, has_gpu, device, numa, dist):
+    """The BASELINE metric's real data plane: one native harness run
+    per rank ("one QP per MI355X") through ibv_reg_mr on the region
+    (peer MR = the rocp2p bridge path) + chained one-sided RDMA WRs.
+    The harness times exactly K steps after W warmup steps inside the
+    native plane; ranks barrier around the run and aggregate as
+    sum(bytes) / max(elapsed), same as the python paths."""
+    import torch
+    from rocnrdma_amd.transport.verbs import run_harness_steps
+
+    mr = args.verbs_mr
+    if mr == "auto":
+        mr = "peer" if has_gpu else "host"
+    dev_idx = device.index if device is not None else 0
+
+    if dist is not None:
+        dist.barrier()
+    rec = run_harness_steps(msg_bytes=msg, region_bytes=region,
+                            steps=args.steps, warmup=args.warmup,
+                            direction=args.direction, mr=mr,
+                            device_index=dev_idx, seed=0xC0FFEE + rank)
+    if dist is not None:
+        dist.barrier()
+
+    elapsed = float(rec["secs"])
+    if dist is not None:
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    msgs_per_step = int(rec["msgs"]) // max(args.steps, 1)
+    total_bytes = int(rec["msgs"]) * msg * world
+    value = total_bytes / elapsed / 1e9
+    integrity = rec.get("integrity", "skipped")
+
+    # metric names 4KB/1MB/64MB: quick per-GPU points outside the
+    # timed region (single-rank runs only, like the sdma path)
+    msg_sweep = None
+    if world == 1:
+        msg_sweep = {}
+        for other in (4096, 1 << 20):
+            if other == msg:
+                continue
+            sweep_region = max(min(region, 64 << 20) // other, 1) * other
+            r2 = run_harness_steps(msg_bytes=other,
+                                   region_bytes=sweep_region,
+                                   steps=1, warmup=1,
+                                   direction=args.direction, mr=mr,
+                                   device_index=dev_idx)
+            msg_sweep[str(other)] = round(float(r2["gbps"]), 3)
+        msg_sweep[str(msg)] = round(value, 3)
+
+    emit_result(args, rank, world, has_gpu, value=value, elapsed=elapsed,
+                msgs_per_step=msgs_per_step, msg=msg, region=region,
+                transport="verbs", integrity=integrity, numa=numa,
+                msg_sweep=msg_sweep, dist=dist, mr=mr)
 
 
 if __name__ == "__main__":

@@ -79,6 +79,29 @@ static Result run_lat(Transport& tp, int iters, uint64_t seed,
   return r;
 }
 
+// bench.py step mode: W untimed warmup steps then EXACTLY K timed
+// steps, one full region pass per step (the driver's steps/warmup
+// contract, executed inside the native data plane).
+static Result run_steps(Transport& tp, int steps, int warmup,
+                        uint64_t seed, bool integrity) {
+  Result r;
+  uint64_t mps = tp.msgs_per_region();
+  for (int w = 0; w < warmup; w++) {
+    tp.post_many((uint64_t)w * mps, mps);
+    tp.flush();
+  }
+  auto t0 = clk::now();
+  for (int s = 0; s < steps; s++) {
+    tp.post_many((uint64_t)s * mps, mps);
+    tp.flush();
+  }
+  r.secs = std::chrono::duration<double>(clk::now() - t0).count();
+  r.msgs = (uint64_t)steps * mps;
+  r.gbps = (double)r.msgs * tp.msg_bytes() / r.secs / 1e9;
+  if (integrity) r.bad = tp.integrity_check(seed);
+  return r;
+}
+
 static Result run_point(Transport& tp, double secs, uint64_t seed,
                         bool integrity) {
   Result r;
@@ -112,6 +135,8 @@ int main(int argc, char** argv) {
   int gpus = 1;
   bool sweep = false, json = false, integrity = true, bidir = false;
   int lat_iters = 0;
+  int steps = 0, step_warmup = 0;
+  int device_index = 0;
   int serve_port = -1;
   bool remote_selftest = false;
   std::string connect_to;
@@ -145,12 +170,38 @@ int main(int argc, char** argv) {
     else if (a == "--connect") connect_to = next();
     else if (a == "--remote-selftest") remote_selftest = true;
     else if (a == "--lat") lat_iters = atoi(next().c_str());
+    else if (a == "--steps") steps = atoi(next().c_str());
+    else if (a == "--warmup") step_warmup = atoi(next().c_str());
+    else if (a == "--device") device_index = atoi(next().c_str());
     else if (a == "--sweep") sweep = true;
     else if (a == "--json") json = true;
     else if (a == "--no-integrity") integrity = false;
     else {
       fprintf(stderr, "unknown arg %s\n", a.c_str());
       return 2;
+    }
+  }
+
+  if (steps > 0) {
+    // bench.py step mode: one transport, one JSON line with the timed
+    // seconds so the caller can aggregate across ranks.
+    try {
+      cfg.device_index = device_index;
+      auto tp = make_transport(transport, cfg);
+      Result r = run_steps(*tp, steps, step_warmup, seed, integrity);
+      const char* ok =
+          !integrity ? "skipped" : (r.bad == 0 ? "ok" : "FAILED");
+      printf("{\"transport\":\"%s\",\"mode\":\"steps\",\"msg_bytes\":%zu,"
+             "\"region_bytes\":%zu,\"mr\":\"%s\",\"steps\":%d,"
+             "\"warmup\":%d,\"secs\":%.6f,\"msgs\":%llu,\"gbps\":%.3f,"
+             "\"integrity\":\"%s\"}\n",
+             tp->name(), cfg.msg_bytes, cfg.region_bytes,
+             cfg.verbs_mr.c_str(), steps, step_warmup, r.secs,
+             (unsigned long long)r.msgs, r.gbps, ok);
+      return (integrity && r.bad) ? 1 : 0;
+    } catch (const std::exception& e) {
+      fprintf(stderr, "steps: %s\n", e.what());
+      return 3;
     }
   }
 

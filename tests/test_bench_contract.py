@@ -94,3 +94,78 @@ def test_msg_sweep_embedded_single_rank():
     assert sw is not None
     assert str(r["config"]["msg_bytes"]) in sw
     assert "4096" in sw and all(v > 0 for v in sw.values())
+
+
+FAKEVERBS_BIN = os.path.join(ROOT, "harness", "build",
+                             "rocp2p_bw_fakeverbs")
+
+
+def test_auto_transport_order():
+    """--transport auto must prefer the real verbs data
+    plane whenever a usable stack exists — never hard-pick sdma on a
+    GPU box."""
+    sys.path.insert(0, ROOT)
+    import bench
+
+    assert bench.choose_transport(True, True) == "verbs"
+    assert bench.choose_transport(False, True) == "verbs"
+    assert bench.choose_transport(True, False) == "sdma"
+    assert bench.choose_transport(False, False) == "fake"
+
+
+@pytest.mark.timeout(300)
+def test_auto_selects_verbs_and_reports_it():
+    """With a verbs stack present (fake-verbs CI layer + forced
+    availability), `--transport auto` must route through the native
+    verbs data plane and emit transport=verbs in the JSON contract."""
+    if not os.path.exists(FAKEVERBS_BIN):
+        pytest.skip("fakeverbs harness not built")
+    env = dict(os.environ)
+    env["PYTHONPATH"] = ROOT + os.pathsep + env.get("PYTHONPATH", "")
+    env["ROCNR_FORCE_VERBS"] = "1"
+    env["ROCNR_VERBS_HARNESS"] = FAKEVERBS_BIN
+    out = subprocess.run(
+        [sys.executable, BENCH, "--gpus", "1", "--steps", "3",
+         "--warmup", "1", "--msg-bytes", "1048576",
+         "--region-bytes", "16777216"],
+        capture_output=True, text=True, timeout=240, env=env, cwd=ROOT)
+    assert out.returncode == 0, out.stderr
+    line = [l for l in out.stdout.splitlines() if l.startswith("{")][-1]
+    r = json.loads(line)
+    assert r["config"]["transport"] == "verbs"
+    assert r["config"]["integrity"] == "ok"
+    assert r["value"] > 0
+    assert r["steps"] == 3 and r["warmup"] == 1
+    # the metric's three sizes are present for single-rank runs
+    assert set(r["config"]["msg_sweep_gbps"]) == {"4096", "1048576"}
+
+
+@pytest.mark.timeout(300)
+def test_two_rank_verbs_aggregate():
+    """Two verbs ranks (fake-verbs layer) over gloo: the whole-job
+    aggregate covers both ranks' native runs, max-elapsed over ranks."""
+    if not os.path.exists(FAKEVERBS_BIN):
+        pytest.skip("fakeverbs harness not built")
+    procs = []
+    port = "29537"
+    for rank in range(2):
+        env = dict(os.environ)
+        env["PYTHONPATH"] = ROOT + os.pathsep + env.get("PYTHONPATH", "")
+        env.update({"RANK": str(rank), "WORLD_SIZE": "2",
+                    "LOCAL_RANK": str(rank), "MASTER_ADDR": "127.0.0.1",
+                    "MASTER_PORT": port, "ROCNR_FORCE_VERBS": "1",
+                    "ROCNR_VERBS_HARNESS": FAKEVERBS_BIN})
+        procs.append(subprocess.Popen(
+            [sys.executable, BENCH, "--gpus", "2", "--msg-bytes", "65536",
+             "--region-bytes", "1048576", "--steps", "3", "--warmup", "1"],
+            stdout=subprocess.PIPE, stderr=subprocess.PIPE, text=True,
+            env=env, cwd=ROOT))
+    outs = [p.communicate(timeout=240) for p in procs]
+    for p, (so, se) in zip(procs, outs):
+        assert p.returncode == 0, se
+    lines = [l for l in outs[0][0].splitlines() if l.startswith("{")]
+    assert len(lines) == 1
+    r = json.loads(lines[0])
+    assert r["config"]["transport"] == "verbs"
+    assert r["n_gpus"] == 2
+    assert "x2" in r["config"]["parallelism"]
