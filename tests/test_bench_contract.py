@@ -169,3 +169,36 @@ def test_two_rank_verbs_aggregate():
     assert r["config"]["transport"] == "verbs"
     assert r["n_gpus"] == 2
     assert "x2" in r["config"]["parallelism"]
+
+
+@pytest.mark.timeout(300)
+def test_eight_rank_gloo_aggregate():
+    """8-rank fan-out math (the 8-GPU node shape) on
+    CPU — whole-job value is sum over ranks of bytes / max elapsed,
+    parallelism records x8."""
+    procs = []
+    port = "29541"
+    for rank in range(8):
+        env = dict(os.environ)
+        env["PYTHONPATH"] = ROOT + os.pathsep + env.get("PYTHONPATH", "")
+        env.update({"RANK": str(rank), "WORLD_SIZE": "8",
+                    "LOCAL_RANK": str(rank), "MASTER_ADDR": "127.0.0.1",
+                    "MASTER_PORT": port})
+        procs.append(subprocess.Popen(
+            [sys.executable, BENCH, "--transport", "fake", "--gpus", "8",
+             "--msg-bytes", "16384", "--region-bytes", "262144",
+             "--steps", "2", "--warmup", "1"],
+            stdout=subprocess.PIPE, stderr=subprocess.PIPE, text=True,
+            env=env, cwd=ROOT))
+    outs = [p.communicate(timeout=240) for p in procs]
+    for p, (so, se) in zip(procs, outs):
+        assert p.returncode == 0, se
+    lines = [l for l in outs[0][0].splitlines() if l.startswith("{")]
+    assert len(lines) == 1
+    r = json.loads(lines[0])
+    assert r["n_gpus"] == 8
+    assert r["config"]["parallelism"] == "1qp-per-gpu x8"
+    # aggregate must be the whole-job sum: 8 ranks x 2 steps x 16 msgs
+    # x 16 KiB over the max elapsed -> global_batch = msgs/step x 8
+    assert r["config"]["global_batch"] == 16 * 8
+    assert r["value"] > 0
