@@ -12,6 +12,8 @@ struct fk_alloc {
 	uint64_t size;
 	unsigned int frag_every;
 	uint64_t bus_base;
+	uint8_t *backing;	/* real memory behind the bus range (or NULL) */
+	uint64_t backing_len;
 };
 
 struct fk_pin {
@@ -39,6 +41,7 @@ static unsigned int fk_bus_skew;
 
 static void fk_destroy_pin_locked(struct fk_pin *p);
 static void fk_tomb_add_locked(const void *info);
+static struct fk_alloc *fk_find_locked(uint64_t addr);
 
 uint64_t fake_kfd_alloc(uint64_t size, unsigned int frag_every)
 {
@@ -60,6 +63,60 @@ uint64_t fake_kfd_alloc(uint64_t size, unsigned int frag_every)
 	fk_allocs = a;
 	pthread_mutex_unlock(&fk_lock);
 	return a->va;
+}
+
+uint64_t fake_kfd_alloc_backed(uint64_t size)
+{
+	struct fk_alloc *a = calloc(1, sizeof(*a));
+	uint64_t nchunks = (size + FAKE_KFD_VRAM_PAGE - 1) / FAKE_KFD_VRAM_PAGE;
+
+	a->backing_len = nchunks * FAKE_KFD_VRAM_PAGE;
+	a->backing = aligned_alloc(4096, a->backing_len);
+	if (!a->backing) {
+		free(a);
+		return 0;
+	}
+	pthread_mutex_lock(&fk_lock);
+	a->va = fk_next_va;
+	a->size = size;
+	a->frag_every = 0;
+	/* the bus range IS the backing memory: bus_to_ptr is identity */
+	a->bus_base = (uint64_t)(uintptr_t)a->backing;
+	fk_next_va += (size + (1ULL << 30)) & ~((1ULL << 21) - 1);
+	a->next = fk_allocs;
+	fk_allocs = a;
+	pthread_mutex_unlock(&fk_lock);
+	return a->va;
+}
+
+void *fake_kfd_bus_to_ptr(uint64_t bus)
+{
+	struct fk_alloc *a;
+	void *p = NULL;
+
+	pthread_mutex_lock(&fk_lock);
+	for (a = fk_allocs; a; a = a->next) {
+		if (a->backing && bus >= a->bus_base &&
+		    bus < a->bus_base + a->backing_len) {
+			p = a->backing + (bus - a->bus_base);
+			break;
+		}
+	}
+	pthread_mutex_unlock(&fk_lock);
+	return p;
+}
+
+uint64_t fake_kfd_backing_bus(uint64_t va)
+{
+	struct fk_alloc *a;
+	uint64_t bus = 0;
+
+	pthread_mutex_lock(&fk_lock);
+	a = fk_find_locked(va);
+	if (a)
+		bus = a->bus_base;
+	pthread_mutex_unlock(&fk_lock);
+	return bus;
 }
 
 static struct fk_alloc *fk_find_locked(uint64_t addr)
@@ -352,6 +409,7 @@ void fake_kfd_free(uint64_t va)
 				}
 			}
 			pthread_mutex_unlock(&fk_lock);
+			free(a->backing);
 			free(a);
 			return;
 		}
@@ -384,6 +442,7 @@ void fake_kfd_reset(void)
 		fk_destroy_pin_locked(p);
 	while ((a = fk_allocs)) {
 		fk_allocs = a->next;
+		free(a->backing);
 		free(a);
 	}
 	fk_stat_get = fk_stat_put = fk_stat_bad_put = fk_stat_cb = 0;

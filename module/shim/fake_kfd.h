@@ -37,6 +37,17 @@ uint64_t fake_kfd_alloc(uint64_t size, unsigned int frag_every);
  * invalidation path), then reclaims them.  Safe from any thread. */
 void fake_kfd_free(uint64_t va);
 
+/* BACKED allocation: like fake_kfd_alloc (contiguous, no frag holes)
+ * but the bus range is real host memory, so mmap-style consumers (the
+ * probe CLI preload loopback) can materialize CPU windows with real
+ * data.  Returns the fake GPU VA. */
+uint64_t fake_kfd_alloc_backed(uint64_t size);
+/* Bus address -> backing pointer for backed allocations (NULL for
+ * synthetic bus ranges). */
+void *fake_kfd_bus_to_ptr(uint64_t bus);
+/* First bus address of a backed allocation's range (0 if va unknown). */
+uint64_t fake_kfd_backing_bus(uint64_t va);
+
 void fake_kfd_reset(void);
 
 /* Introspection for tests. */

@@ -3,7 +3,11 @@
 // kernels (p2p_kernels.hip), the C++ harness backends and the probe
 // CLI.  Must stay bit-identical to rocnrdma_amd/utils/pattern.py.
 #pragma once
+#ifdef __cplusplus
 #include <cstdint>
+#else
+#include <stdint.h>
+#endif
 
 #if defined(__HIPCC__) || defined(__HIP_DEVICE_COMPILE__)
 #define ROCP2P_HD __host__ __device__ inline

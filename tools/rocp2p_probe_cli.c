@@ -34,6 +34,7 @@
 #include <sys/mman.h>
 
 #include "rocp2p_probe_abi.h"
+#include "../rocnrdma_amd/ops/csrc/p2p_pattern.h"
 
 #ifdef WITH_HIP
 #include <hip/hip_runtime.h>
@@ -102,6 +103,55 @@ static int cmd_info(uint64_t addr, uint64_t len)
 	       (unsigned long long)p.total_bytes,
 	       (unsigned long long)p.first_dma_addr,
 	       (unsigned long long)p.max_seg_bytes);
+	return 0;
+}
+
+/* Full flow on a CALLER-SUPPLIED range (no GPU needed): pin, duplicate
+ * pin, info, page size, mmap CPU window, optional splitmix64 pattern
+ * verification (the preload loopback patterns the backing with the
+ * same generator — rocnrdma_amd/ops/csrc/p2p_pattern.h), unpin-all.
+ * This is the flow CI executes through the UNMODIFIED binary via
+ * module/shim/probe_preload.c (LD_PRELOAD). */
+static int cmd_selftest_extern(uint64_t addr, uint64_t len, uint64_t seed,
+			       int have_seed)
+{
+	struct rocp2p_probe_pin pin = { .addr = addr, .length = len };
+	struct rocp2p_probe_unpin unpin = { .addr = addr, .length = len };
+	struct rocp2p_probe_info info = { .addr = addr, .length = len };
+	struct rocp2p_probe_page_size ps = { .addr = addr, .length = len };
+	uint64_t *map, i, words = len / 8, bad = 0;
+
+	need(cmd_is_gpu(addr) == 0, "address classified");
+	need(ioctl(dev_fd, ROCP2P_PROBE_GET_PAGE_SIZE, &ps) == 0,
+	     "GET_PAGE_SIZE");
+	printf("page_size %llu\n", (unsigned long long)ps.page_size);
+
+	need(ioctl(dev_fd, ROCP2P_PROBE_GET_PAGES, &pin) == 0, "pin #1");
+	need(ioctl(dev_fd, ROCP2P_PROBE_GET_PAGES, &pin) == 0,
+	     "pin #2 (duplicate range)");
+
+	need(ioctl(dev_fd, ROCP2P_PROBE_GET_INFO, &info) == 0, "GET_INFO");
+	printf("nents %llu total %llu max_seg %llu\n",
+	       (unsigned long long)info.nents,
+	       (unsigned long long)info.total_bytes,
+	       (unsigned long long)info.max_seg_bytes);
+	need(info.total_bytes == len, "sg covers the pin");
+
+	map = (uint64_t *)mmap(NULL, len, PROT_READ, MAP_SHARED, dev_fd,
+			       (off_t)addr);
+	need(map != MAP_FAILED, "mmap CPU window");
+	if (have_seed) {
+		for (i = 0; i < words; i++)
+			bad += (map[i] != rocp2p_pattern_word(seed, i));
+		need(bad == 0, "mmap readback matches pattern");
+		printf("mmap readback: %llu words OK\n",
+		       (unsigned long long)words);
+	}
+	munmap(map, len);
+
+	need(ioctl(dev_fd, ROCP2P_PROBE_PUT_PAGES, &unpin) == 0, "unpin");
+	need(unpin.released == 2, "both duplicate pins released");
+	printf("SELFTEST-EXTERN PASSED\n");
 	return 0;
 }
 
@@ -199,6 +249,12 @@ int main(int argc, char **argv)
 	if (!strcmp(argv[1], "info") && argc == 4)
 		return cmd_info(strtoull(argv[2], 0, 16),
 				strtoull(argv[3], 0, 0));
+	if (!strcmp(argv[1], "selftest-extern") && (argc == 4 || argc == 5))
+		return cmd_selftest_extern(strtoull(argv[2], 0, 16),
+					   strtoull(argv[3], 0, 0),
+					   argc == 5 ?
+						strtoull(argv[4], 0, 0) : 0,
+					   argc == 5);
 #ifdef WITH_HIP
 	if (!strcmp(argv[1], "selftest"))
 		return cmd_selftest(argc > 2 ? strtoull(argv[2], 0, 0) : 64);
