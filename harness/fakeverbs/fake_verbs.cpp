@@ -20,6 +20,13 @@
 #include <deque>
 #include <vector>
 
+#ifdef ROCNR_FAKEVERBS_PEER
+// Full-stack build: ibv_reg_mr on a fake-GPU VA dispatches through the
+// REAL rocp2p bridge (userspace shim) the way the IB core's peer-
+// memory probe does — see module/shim/peer_glue.h for the call stack.
+#include "../../module/shim/peer_glue.h"
+#endif
+
 namespace {
 
 struct FakeCtx;
@@ -34,6 +41,11 @@ struct FakeMr {
   // translated iova -> host_map.
   uint8_t* host_map = nullptr;
   size_t map_len = 0;
+#ifdef ROCNR_FAKEVERBS_PEER
+  // peer MRs: handle + device-mapped sg table from the real bridge
+  void* peer_handle = nullptr;
+  std::vector<rocnr_glue_seg> segs;
+#endif
 };
 
 struct FakeCq {
@@ -77,9 +89,21 @@ bool range_ok(const FakeMr* m, uint64_t addr, uint64_t len) {
 }
 
 // Data-plane pointer for an in-MR address: identity for host MRs,
-// iova->CPU-window translation for dmabuf MRs.
+// iova->CPU-window translation for dmabuf MRs, sg-walk through the
+// bridge's device-mapped table for peer MRs.
 uint8_t* mr_data_ptr(const FakeMr* m, uint64_t addr) {
   if (m->host_map) return m->host_map + (addr - (uint64_t)m->mr.addr);
+#ifdef ROCNR_FAKEVERBS_PEER
+  if (!m->segs.empty()) {
+    uint64_t off = addr - (uint64_t)m->mr.addr;
+    for (const auto& s : m->segs) {
+      if (off < s.len)
+        return (uint8_t*)rocnr_glue_bus_ptr(s.bus + off);
+      off -= s.len;
+    }
+    return nullptr;
+  }
+#endif
   return (uint8_t*)addr;
 }
 
@@ -111,6 +135,26 @@ struct ibv_mr* ibv_reg_mr(struct ibv_pd* pd, void* addr, size_t length,
                           int access) {
   if (!pd || !addr || !length) return nullptr;
   auto* m = new FakeMr();
+#ifdef ROCNR_FAKEVERBS_PEER
+  // The IB core's peer-memory probe: a VA that normal pinning cannot
+  // claim is offered to registered peer clients — here the REAL
+  // rocp2p bridge, which pins through the (fake) KFD and returns the
+  // device-mapped sg table the HCA would DMA against.
+  if (rocnr_glue_init() == 0 && rocnr_glue_is_gpu((uint64_t)addr)) {
+    rocnr_glue_seg segs[512];
+    int n = 512;
+    void* h = nullptr;
+    int r = rocnr_glue_reg_mr((uint64_t)addr, length, &h, segs, &n);
+    if (r != 0) {
+      fprintf(stderr, "fake_verbs: peer-memory registration failed (%d)\n",
+              r);
+      delete m;
+      return nullptr;
+    }
+    m->peer_handle = h;
+    m->segs.assign(segs, segs + n);
+  }
+#endif
   m->mr.pd = pd;
   m->mr.addr = addr;
   m->mr.length = length;
@@ -159,6 +203,15 @@ int ibv_dereg_mr(struct ibv_mr* mr) {
       if (m->host_map)
         munmap(m->host_map - (m->map_len - m->mr.length), m->map_len);
       m->host_map = nullptr;
+#ifdef ROCNR_FAKEVERBS_PEER
+      if (m->peer_handle) {
+        // ibv_dereg_mr path: dma_unmap -> put_pages -> release through
+        // the real bridge (fake IB core teardown ordering)
+        rocnr_glue_dereg_mr(m->peer_handle);
+        m->peer_handle = nullptr;
+        m->segs.clear();
+      }
+#endif
       m->live = false;
       return 0;
     }

@@ -76,6 +76,9 @@ bool verbs_runtime_available() {
 #include "rocp2p_oob.h"
 #include "../../rocnrdma_amd/ops/csrc/p2p_kernels.h"
 #include "../../rocnrdma_amd/ops/csrc/p2p_pattern.h"
+#ifdef ROCNR_FAKEVERBS_PEER
+#include "../../module/shim/peer_glue.h"
+#endif
 
 namespace rocp2p {
 
@@ -213,6 +216,7 @@ struct Region {
   ibv_mr* mr = nullptr;
   size_t bytes = 0;
   bool host_is_map = false;  // host came from mmap (memfd dmabuf stand-in)
+  uint64_t peer_va = 0;      // full-stack build: glue-owned fake VRAM
   std::string mode;
 
   void create(ibv_pd* pd, size_t nbytes, std::string mr_mode,
@@ -263,6 +267,21 @@ struct Region {
       return;
     }
     // peer mode: dispatched to the rocp2p bridge by the IB core
+#ifdef ROCNR_FAKEVERBS_PEER
+    if (!hip_available()) {
+      // full-stack CI build: "VRAM" is a backed fake-KFD allocation;
+      // ibv_reg_mr(va) below dispatches through the REAL bridge
+      VB_THROW(rocnr_glue_init() == 0, "peer glue init");
+      peer_va = rocnr_glue_alloc(nbytes);
+      VB_THROW(peer_va, "glue VRAM alloc failed");
+      ptr = (uint8_t*)peer_va;
+      host = (uint8_t*)rocnr_glue_vram_ptr(peer_va);  // CPU view of VRAM
+      VB_THROW(host, "glue VRAM ptr");
+      mr = ibv_reg_mr(pd, ptr, nbytes, acc);
+      VB_THROW(mr, "ibv_reg_mr(GPU VA) failed — peer client not dispatched?");
+      return;
+    }
+#endif
     VB_THROW(hipSetDevice(device_index) == hipSuccess, "hipSetDevice");
     VB_THROW(hipMalloc((void**)&gpu, nbytes) == hipSuccess,
              "hipMalloc region");
@@ -305,13 +324,18 @@ struct Region {
   void destroy() {
     if (mr) ibv_dereg_mr(mr);
     if (host) {
-      if (host_is_map) munmap(host, bytes);
+      if (peer_va) { /* backing owned by the glue */ }
+      else if (host_is_map) munmap(host, bytes);
       else free(host);
     }
+#ifdef ROCNR_FAKEVERBS_PEER
+    if (peer_va) rocnr_glue_free(peer_va);
+#endif
     if (gpu) (void)hipFree(gpu);
     mr = nullptr;
     host = gpu = ptr = nullptr;
     host_is_map = false;
+    peer_va = 0;
   }
 };
 

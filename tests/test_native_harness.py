@@ -183,3 +183,62 @@ def test_verbs_wr_chaining(built_fakeverbs, chain):
     r = json.loads(out.stdout.strip())
     assert r["integrity"] == "ok"
     assert r["msgs_per_s"] > 0
+
+
+FULLSTACK = os.path.join(HARNESS, "build", "rocp2p_bw_fullstack")
+
+
+@pytest.fixture(scope="module")
+def built_fullstack(built):
+    assert os.path.exists(FULLSTACK)
+    return FULLSTACK
+
+
+@pytest.mark.parametrize("direction", ["write", "read"])
+@pytest.mark.timeout(300)
+def test_full_stack_peer_mr(built_fullstack, direction):
+    """The reference's COMPLETE L5->L0 flow in one
+    process: ibv_reg_mr on a GPU VA -> IB-core peer probe -> the REAL
+    rocp2p bridge (acquire/get_pages/dma_map against the fake KFD) ->
+    one-sided DMA against the bridge's device-mapped sg table ->
+    payload verified in 'VRAM'.  No layer mocked out of the control
+    path: the bridge code on this hot path is the same file kbuild
+    compiles (module/bridge/rocp2p_main.c)."""
+    out = subprocess.run(
+        [built_fullstack, "--transport", "verbs", "--mr", "peer",
+         "--msg", "65536", "--region", "8388608", "--secs", "0.1",
+         "--dir", direction, "--json"],
+        capture_output=True, text=True, timeout=120)
+    assert out.returncode == 0, out.stdout + out.stderr
+    r = json.loads(out.stdout.strip().splitlines()[-1])
+    assert r["integrity"] == "ok"
+    # the real bridge registered (its init line proves dispatch)
+    assert "PeerDirect client 'rocp2p'" in out.stderr
+
+
+@pytest.mark.timeout(300)
+def test_full_stack_peer_mr_client_server(built_fullstack):
+    """ib_write_bw server/client shape with the server's region
+    registered through the real bridge (peer MR), client RDMA-writes
+    over TCP-bootstrapped QPs, REMOTE integrity verification."""
+    out = subprocess.run(
+        [built_fullstack, "--remote-selftest", "--mr", "peer",
+         "--msg", "65536", "--region", "8388608", "--secs", "0.1"],
+        capture_output=True, text=True, timeout=120)
+    assert out.returncode == 0, out.stdout + out.stderr
+    r = json.loads(out.stdout.strip().splitlines()[-1])
+    assert r["remote_integrity"] == "ok"
+
+
+@pytest.mark.timeout(300)
+def test_full_stack_peer_mr_chained(built_fullstack):
+    """Chained posting against a bridge-registered MR keeps exact
+    delivery."""
+    out = subprocess.run(
+        [built_fullstack, "--transport", "verbs", "--mr", "peer",
+         "--msg", "4096", "--region", "4194304", "--secs", "0.1",
+         "--chain", "16", "--json"],
+        capture_output=True, text=True, timeout=120)
+    assert out.returncode == 0, out.stdout + out.stderr
+    r = json.loads(out.stdout.strip().splitlines()[-1])
+    assert r["integrity"] == "ok"
