@@ -30,9 +30,12 @@ def available_transports() -> list[str]:
 
 def get_transport(name: str, **kw) -> Transport:
     if name == "auto":
+        # auto picks the best IN-PROCESS plane (sdma/fake).  The verbs
+        # data plane lives in the native harness and is routed by
+        # bench.py (choose_transport / run_verbs) — constructing
+        # VerbsTransport here would raise by design.
         avail = available_transports()
-        name = "verbs" if "verbs" in avail else (
-            "sdma" if "sdma" in avail else "fake")
+        name = "sdma" if "sdma" in avail else "fake"
     if name == "fake":
         from .fake import FakeTransport
 
