@@ -209,8 +209,10 @@ int main(int argc, char** argv) {
     // passive target (ib_write_bw server shape); blocks until the
     // client says bye
     try {
-      run_verbs_target(cfg, serve_port,
-                       [](int p) { printf("listening on port %d\n", p); });
+      run_verbs_target(cfg, serve_port, [](int p) {
+        printf("listening on port %d\n", p);
+        fflush(stdout);  // piped consumers (tests) wait on this line
+      });
     } catch (const std::exception& e) {
       fprintf(stderr, "serve: %s\n", e.what());
       return 3;
