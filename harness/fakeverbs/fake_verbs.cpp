@@ -9,11 +9,21 @@
 // Env knobs:
 //   FAKE_VERBS_LINK=ib|eth   link layer reported by query_port
 //                            (default ib) — covers both AH paths.
+//   FAKE_VERBS_SHM=<name>    cross-process fabric: processes sharing
+//                            the name publish their MRs in a shm
+//                            registry and run a NIC-role engine thread
+//                            that applies one-sided ops addressed to
+//                            their rkeys — two REAL processes move
+//                            real bytes (RDMA semantics: the remote
+//                            application thread is never involved).
 #include "infiniband/verbs.h"
 
+#include <fcntl.h>
+#include <pthread.h>
 #include <sys/mman.h>
 #include <unistd.h>
 
+#include <atomic>
 #include <cstdlib>
 #include <cstring>
 #include <cstdio>
@@ -71,21 +81,233 @@ char g_device_storage;
 ibv_device* g_device_list[2] = {
     reinterpret_cast<ibv_device*>(&g_device_storage), nullptr};
 
+// ---- cross-process fabric (FAKE_VERBS_SHM) --------------------------
+// Layout in one shm segment: an MR registry (rkey -> owner node) and
+// one op ring per node.  A posting process that does not own the
+// target rkey writes a descriptor (+ payload for WRITE) into the
+// OWNER's ring and spins for completion; each process's engine thread
+// services its own ring by resolving the rkey against its local MRs
+// (so peer- and dmabuf-backed regions work as remote targets too).
+namespace fabric {
+
+constexpr uint32_t kMagic = 0xFAB51C05;
+constexpr int kMaxMrs = 256;
+constexpr int kMaxNodes = 4;
+constexpr int kSlots = 4;
+constexpr size_t kSlotPayload = 1u << 20;
+
+enum SlotState : uint32_t { FREE = 0, CLAIMED, POSTED, DONE, ERR };
+
+struct MrEntry {
+  std::atomic<uint32_t> live;
+  uint32_t rkey;
+  uint32_t node;
+  int access;
+  uint64_t iova;
+  uint64_t len;
+};
+
+struct Slot {
+  std::atomic<uint32_t> state;
+  uint32_t op;  // IBV_WR_RDMA_WRITE / IBV_WR_RDMA_READ
+  uint32_t rkey;
+  uint32_t len;
+  uint64_t remote_addr;
+  uint8_t payload[kSlotPayload];
+};
+
+struct Ring {
+  Slot slots[kSlots];
+};
+
+struct Shared {
+  std::atomic<uint32_t> magic;
+  std::atomic<uint32_t> next_node;
+  std::atomic<uint32_t> next_key;  // fabric-global so rkeys never clash
+  MrEntry mrs[kMaxMrs];
+  Ring rings[kMaxNodes];
+};
+
+Shared* g_sh = nullptr;
+int g_node = -1;
+pthread_t g_engine;
+std::atomic<bool> g_engine_stop{false};
+
+uint8_t* local_ptr_for(uint32_t rkey, uint64_t addr, uint64_t len,
+                       int* access_out);
+
+void* engine_main(void*) {
+  Ring& ring = g_sh->rings[g_node];
+  int idle_rounds = 0;
+  while (!g_engine_stop.load(std::memory_order_relaxed)) {
+    bool idle = true;
+    for (auto& s : ring.slots) {
+      if (s.state.load(std::memory_order_acquire) != POSTED) continue;
+      idle = false;
+      int access = 0;
+      uint8_t* p = local_ptr_for(s.rkey, s.remote_addr, s.len, &access);
+      uint32_t next = DONE;
+      if (!p) {
+        next = ERR;
+      } else if (s.op == IBV_WR_RDMA_WRITE) {
+        if (!(access & IBV_ACCESS_REMOTE_WRITE)) next = ERR;
+        else memcpy(p, s.payload, s.len);
+      } else {
+        if (!(access & IBV_ACCESS_REMOTE_READ)) next = ERR;
+        else memcpy(s.payload, p, s.len);
+      }
+      s.state.store(next, std::memory_order_release);
+    }
+    if (idle) {
+      // hybrid poll: stay hot while traffic flows, back off when idle
+      if (++idle_rounds < 5000) sched_yield();
+      else usleep(50);
+    } else {
+      idle_rounds = 0;
+    }
+  }
+  return nullptr;
+}
+
+bool init_from_env() {
+  const char* name = getenv("FAKE_VERBS_SHM");
+  if (!name || g_sh) return g_sh != nullptr;
+  int fd = shm_open(name, O_RDWR | O_CREAT, 0600);
+  if (fd < 0) return false;
+  if (ftruncate(fd, sizeof(Shared)) != 0) {
+    close(fd);
+    return false;
+  }
+  void* m = mmap(nullptr, sizeof(Shared), PROT_READ | PROT_WRITE,
+                 MAP_SHARED, fd, 0);
+  close(fd);
+  if (m == MAP_FAILED) return false;
+  g_sh = reinterpret_cast<Shared*>(m);
+  uint32_t expect = 0;
+  if (g_sh->magic.compare_exchange_strong(expect, 1)) {
+    // first process initializes
+    g_sh->next_node.store(0);
+    g_sh->next_key.store(0x1000);
+    for (auto& e : g_sh->mrs) e.live.store(0);
+    for (auto& r : g_sh->rings)
+      for (auto& s : r.slots) s.state.store(FREE);
+    g_sh->magic.store(kMagic, std::memory_order_release);
+  } else {
+    while (g_sh->magic.load(std::memory_order_acquire) != kMagic)
+      usleep(100);
+  }
+  g_node = (int)g_sh->next_node.fetch_add(1);
+  if (g_node >= kMaxNodes) {
+    fprintf(stderr, "fake_verbs: fabric node limit\n");
+    g_sh = nullptr;
+    return false;
+  }
+  g_engine_stop.store(false);
+  pthread_create(&g_engine, nullptr, engine_main, nullptr);
+  return true;
+}
+
+void publish(uint32_t rkey, uint64_t iova, uint64_t len, int access) {
+  if (!g_sh) return;
+  for (auto& e : g_sh->mrs) {
+    uint32_t expect = 0;
+    if (e.live.compare_exchange_strong(expect, 1)) {
+      e.rkey = rkey;
+      e.node = (uint32_t)g_node;
+      e.iova = iova;
+      e.len = len;
+      e.access = access;
+      e.live.store(2, std::memory_order_release);
+      return;
+    }
+  }
+  fprintf(stderr, "fake_verbs: fabric MR registry full\n");
+}
+
+void unpublish(uint32_t rkey) {
+  if (!g_sh) return;
+  for (auto& e : g_sh->mrs)
+    if (e.live.load(std::memory_order_acquire) == 2 && e.rkey == rkey &&
+        e.node == (uint32_t)g_node)
+      e.live.store(0, std::memory_order_release);
+}
+
+const MrEntry* lookup(uint32_t rkey) {
+  if (!g_sh) return nullptr;
+  for (auto& e : g_sh->mrs)
+    if (e.live.load(std::memory_order_acquire) == 2 && e.rkey == rkey)
+      return &e;
+  return nullptr;
+}
+
+// One remote one-sided op (chunks > slot payload are split by caller).
+int remote_op(const MrEntry* mr, uint32_t op, uint64_t remote_addr,
+              uint8_t* local, uint32_t len) {
+  Ring& ring = g_sh->rings[mr->node];
+  for (;;) {
+    for (auto& s : ring.slots) {
+      uint32_t expect = FREE;
+      if (!s.state.compare_exchange_strong(expect, CLAIMED)) continue;
+      s.op = op;
+      s.rkey = mr->rkey;
+      s.len = len;
+      s.remote_addr = remote_addr;
+      if (op == IBV_WR_RDMA_WRITE) memcpy(s.payload, local, len);
+      s.state.store(POSTED, std::memory_order_release);
+      // wait for the owner's engine: hot spin first, then 50 us polls
+      // (bounded ~10 s total)
+      for (long spins = 0; spins < 300000; spins++) {
+        uint32_t st = s.state.load(std::memory_order_acquire);
+        if (st == DONE || st == ERR) {
+          int rc = (st == DONE) ? 0 : -1;
+          if (!rc && op == IBV_WR_RDMA_READ)
+            memcpy(local, s.payload, len);
+          s.state.store(FREE, std::memory_order_release);
+          return rc;
+        }
+        if (spins < 100000) sched_yield();
+        else usleep(50);
+      }
+      s.state.store(FREE, std::memory_order_release);
+      fprintf(stderr, "fake_verbs: fabric op timed out\n");
+      return -1;
+    }
+    usleep(50);
+  }
+}
+
+}  // namespace fabric
+
 #define FAIL(msg)                                          \
   do {                                                     \
     fprintf(stderr, "fake_verbs: %s\n", msg);              \
     return -1;                                             \
   } while (0)
 
+// g_ctx.mrs is shared with the fabric engine thread.
+pthread_mutex_t g_mr_mu = PTHREAD_MUTEX_INITIALIZER;
+
 FakeMr* find_mr_by_key(uint32_t key, bool remote) {
+  pthread_mutex_lock(&g_mr_mu);
   for (auto* m : g_ctx.mrs)
-    if (m->live && (remote ? m->mr.rkey : m->mr.lkey) == key) return m;
+    if (m->live && (remote ? m->mr.rkey : m->mr.lkey) == key) {
+      pthread_mutex_unlock(&g_mr_mu);
+      return m;
+    }
+  pthread_mutex_unlock(&g_mr_mu);
   return nullptr;
 }
 
 bool range_ok(const FakeMr* m, uint64_t addr, uint64_t len) {
   uint64_t base = (uint64_t)m->mr.addr;
   return addr >= base && addr + len <= base + m->mr.length;
+}
+
+// rkeys/lkeys: fabric-global when the cross-process fabric is up (so
+// two processes never mint the same key), else process-local.
+uint32_t next_key() {
+  if (fabric::g_sh) return fabric::g_sh->next_key.fetch_add(1);
+  return g_ctx.next_key++;
 }
 
 // Data-plane pointer for an in-MR address: identity for host MRs,
@@ -109,6 +331,20 @@ uint8_t* mr_data_ptr(const FakeMr* m, uint64_t addr) {
 
 }  // namespace
 
+namespace {
+namespace fabric {
+// Engine-side rkey resolution against this process's MRs (any MR kind:
+// host, dmabuf window, bridge-registered peer).
+uint8_t* local_ptr_for(uint32_t rkey, uint64_t addr, uint64_t len,
+                       int* access_out) {
+  FakeMr* m = find_mr_by_key(rkey, true);
+  if (!m || !range_ok(m, addr, len)) return nullptr;
+  *access_out = m->access;
+  return mr_data_ptr(m, addr);
+}
+}  // namespace fabric
+}  // namespace
+
 extern "C" {
 
 struct ibv_device** ibv_get_device_list(int* num) {
@@ -119,8 +355,9 @@ struct ibv_device** ibv_get_device_list(int* num) {
 void ibv_free_device_list(struct ibv_device**) {}
 
 struct ibv_context* ibv_open_device(struct ibv_device* d) {
-  return d == g_device_list[0] ? reinterpret_cast<ibv_context*>(&g_ctx)
-                               : nullptr;
+  if (d != g_device_list[0]) return nullptr;
+  fabric::init_from_env();  // no-op unless FAKE_VERBS_SHM is set
+  return reinterpret_cast<ibv_context*>(&g_ctx);
 }
 
 int ibv_close_device(struct ibv_context*) { return 0; }
@@ -158,11 +395,14 @@ struct ibv_mr* ibv_reg_mr(struct ibv_pd* pd, void* addr, size_t length,
   m->mr.pd = pd;
   m->mr.addr = addr;
   m->mr.length = length;
-  m->mr.lkey = g_ctx.next_key++;
-  m->mr.rkey = g_ctx.next_key++;
+  m->mr.lkey = next_key();
+  m->mr.rkey = next_key();
   m->access = access;
   m->live = true;
+  pthread_mutex_lock(&g_mr_mu);
   g_ctx.mrs.push_back(m);
+  pthread_mutex_unlock(&g_mr_mu);
+  fabric::publish(m->mr.rkey, (uint64_t)addr, length, access);
   return &m->mr;
 }
 
@@ -187,35 +427,45 @@ struct ibv_mr* ibv_reg_dmabuf_mr(struct ibv_pd* pd, uint64_t offset,
   m->mr.pd = pd;
   m->mr.addr = (void*)iova;
   m->mr.length = length;
-  m->mr.lkey = g_ctx.next_key++;
-  m->mr.rkey = g_ctx.next_key++;
+  m->mr.lkey = next_key();
+  m->mr.rkey = next_key();
   m->access = access;
   m->live = true;
   m->host_map = (uint8_t*)map + offset;
   m->map_len = length + offset;
+  pthread_mutex_lock(&g_mr_mu);
   g_ctx.mrs.push_back(m);
+  pthread_mutex_unlock(&g_mr_mu);
+  fabric::publish(m->mr.rkey, iova, length, access);
   return &m->mr;
 }
 
 int ibv_dereg_mr(struct ibv_mr* mr) {
+  fabric::unpublish(mr->rkey);
+  FakeMr* found = nullptr;
+  pthread_mutex_lock(&g_mr_mu);
   for (auto* m : g_ctx.mrs)
     if (&m->mr == mr && m->live) {
-      if (m->host_map)
-        munmap(m->host_map - (m->map_len - m->mr.length), m->map_len);
-      m->host_map = nullptr;
-#ifdef ROCNR_FAKEVERBS_PEER
-      if (m->peer_handle) {
-        // ibv_dereg_mr path: dma_unmap -> put_pages -> release through
-        // the real bridge (fake IB core teardown ordering)
-        rocnr_glue_dereg_mr(m->peer_handle);
-        m->peer_handle = nullptr;
-        m->segs.clear();
-      }
-#endif
       m->live = false;
-      return 0;
+      found = m;
+      break;
     }
-  return -1;
+  pthread_mutex_unlock(&g_mr_mu);
+  if (!found) return -1;
+  if (found->host_map)
+    munmap(found->host_map - (found->map_len - found->mr.length),
+           found->map_len);
+  found->host_map = nullptr;
+#ifdef ROCNR_FAKEVERBS_PEER
+  if (found->peer_handle) {
+    // ibv_dereg_mr path: dma_unmap -> put_pages -> release through
+    // the real bridge (fake IB core teardown ordering)
+    rocnr_glue_dereg_mr(found->peer_handle);
+    found->peer_handle = nullptr;
+    found->segs.clear();
+  }
+#endif
+  return 0;
 }
 
 struct ibv_cq* ibv_create_cq(struct ibv_context*, int cqe, void*, void*,
@@ -333,24 +583,45 @@ int ibv_post_send(struct ibv_qp* qp, struct ibv_send_wr* wr,
     if (!local) FAIL("post_send: bad lkey");
     if (!range_ok(local, wr->sg_list[0].addr, wr->sg_list[0].length))
       FAIL("post_send: local sge out of MR bounds");
-    FakeMr* remote = find_mr_by_key(wr->wr.rdma.rkey, true);
-    if (!remote) FAIL("post_send: bad rkey");
-    if (!range_ok(remote, wr->wr.rdma.remote_addr, wr->sg_list[0].length))
-      FAIL("post_send: remote range out of MR bounds");
-    uint8_t* rptr = mr_data_ptr(remote, wr->wr.rdma.remote_addr);
-    uint8_t* lptr = mr_data_ptr(local, wr->sg_list[0].addr);
-    if (wr->opcode == IBV_WR_RDMA_WRITE) {
-      if (!(remote->access & IBV_ACCESS_REMOTE_WRITE))
-        FAIL("post_send: remote MR lacks REMOTE_WRITE");
-      memcpy(rptr, lptr, wr->sg_list[0].length);
-    } else if (wr->opcode == IBV_WR_RDMA_READ) {
-      if (!(remote->access & IBV_ACCESS_REMOTE_READ))
-        FAIL("post_send: remote MR lacks REMOTE_READ");
-      if (!(local->access & IBV_ACCESS_LOCAL_WRITE))
-        FAIL("post_send: local MR lacks LOCAL_WRITE");
-      memcpy(lptr, rptr, wr->sg_list[0].length);
-    } else {
+    if (wr->opcode != IBV_WR_RDMA_WRITE && wr->opcode != IBV_WR_RDMA_READ)
       FAIL("post_send: unsupported opcode");
+    uint8_t* lptr = mr_data_ptr(local, wr->sg_list[0].addr);
+    if (wr->opcode == IBV_WR_RDMA_READ &&
+        !(local->access & IBV_ACCESS_LOCAL_WRITE))
+      FAIL("post_send: local MR lacks LOCAL_WRITE");
+    FakeMr* remote = find_mr_by_key(wr->wr.rdma.rkey, true);
+    if (remote) {
+      if (!range_ok(remote, wr->wr.rdma.remote_addr, wr->sg_list[0].length))
+        FAIL("post_send: remote range out of MR bounds");
+      uint8_t* rptr = mr_data_ptr(remote, wr->wr.rdma.remote_addr);
+      if (wr->opcode == IBV_WR_RDMA_WRITE) {
+        if (!(remote->access & IBV_ACCESS_REMOTE_WRITE))
+          FAIL("post_send: remote MR lacks REMOTE_WRITE");
+        memcpy(rptr, lptr, wr->sg_list[0].length);
+      } else {
+        if (!(remote->access & IBV_ACCESS_REMOTE_READ))
+          FAIL("post_send: remote MR lacks REMOTE_READ");
+        memcpy(lptr, rptr, wr->sg_list[0].length);
+      }
+    } else {
+      // Cross-process fabric: the rkey may belong to another node —
+      // the op is shipped to the owner's NIC-role engine (the remote
+      // application thread is never involved, RDMA-style).
+      const fabric::MrEntry* fe = fabric::lookup(wr->wr.rdma.rkey);
+      if (!fe) FAIL("post_send: bad rkey");
+      uint64_t raddr = wr->wr.rdma.remote_addr;
+      uint32_t len = wr->sg_list[0].length;
+      if (raddr < fe->iova || raddr + len > fe->iova + fe->len)
+        FAIL("post_send: remote range out of fabric MR bounds");
+      for (uint32_t off = 0; off < len;) {
+        uint32_t chunk = len - off;
+        if (chunk > fabric::kSlotPayload)
+          chunk = (uint32_t)fabric::kSlotPayload;
+        if (fabric::remote_op(fe, wr->opcode, raddr + off, lptr + off,
+                              chunk) != 0)
+          FAIL("post_send: fabric remote op failed");
+        off += chunk;
+      }
     }
     if (wr->send_flags & IBV_SEND_SIGNALED) {
       ibv_wc wc;
