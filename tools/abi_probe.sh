@@ -46,9 +46,16 @@ else
         echo "  check $PEER manually."
         rc=1
     fi
-    grep -q 'PEER_MEM_INVALIDATE_UNMAPS' "$PEER" && \
-        echo "  note: header exposes PEER_MEM_INVALIDATE_UNMAPS (informational)"
+    if grep -q 'peer_memory_client_ex' "$PEER"; then
+        HAS_EX=1
+        echo "  extended registration (peer_memory_client_ex): present"
+    else
+        HAS_EX=0
+        echo "  extended registration: ABSENT — build with"
+        echo "  ROCNR_PEER_MEM_HAS_EX=0 (bridge registers the plain client)"
+    fi
 fi
+[ -z "$PEER" ] && HAS_EX=1
 
 echo "== amdkfd RDMA ABI (amd_rdma.h) =="
 if [ -z "$AMDR" ]; then
@@ -76,8 +83,6 @@ echo "== build line =="
 LINE="make -C module/bridge KDIR=/lib/modules/\$(uname -r)/build"
 [ -n "$PEER" ] && LINE="$LINE OFA_DIR=$(dirname "$(dirname "$(dirname "$PEER")")")"
 [ -n "$AMDR" ] && LINE="$LINE AMD_RDMA=$(dirname "$AMDR")"
+LINE="$LINE ROCNR_ABI_FLAGS=\"-DROCNR_PEER_MEM_CORE_CONTEXT_U64=$CORE_U64 -DROCNR_PEER_MEM_HAS_EX=$HAS_EX -DROCNR_AMD_RDMA_HAS_DMA_DEV=$HAS_DMADEV\""
 echo "  $LINE"
-echo "  (vendored-header fallback switches:"
-echo "   ROCNR_PEER_MEM_CORE_CONTEXT_U64=$CORE_U64"
-echo "   ROCNR_AMD_RDMA_HAS_DMA_DEV=$HAS_DMADEV)"
 exit $rc
