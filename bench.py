@@ -273,10 +273,41 @@ def run_verbs(args, msg, region, rank, world, has_gpu, device, numa, dist):
     import torch
     from rocnrdma_amd.transport.verbs import run_harness_steps
 
-    mr = args.verbs_mr
-    if mr == "auto":
-        mr = "peer" if has_gpu else "host"
     dev_idx = device.index if device is not None else 0
+
+    # Pick the MR mode BEFORE the ranks synchronize: peer (the bridge
+    # path, the product under test) is preferred on GPU boxes, but a
+    # box with an HCA and no rocp2p.ko loaded should still measure the
+    # NIC<->HBM metric through the module-free dmabuf path rather than
+    # fail.  A 1-step probe run per candidate keeps every rank on the
+    # same number of barrier calls.
+    if args.verbs_mr != "auto":
+        candidates = [args.verbs_mr]
+    elif has_gpu:
+        candidates = ["peer", "dmabuf"]
+    else:
+        candidates = ["host"]
+    mr = None
+    probe_err = ""
+    for cand in candidates:
+        if len(candidates) == 1:
+            mr = cand
+            break
+        try:
+            run_harness_steps(msg_bytes=min(msg, 1 << 20),
+                              region_bytes=min(region, 16 << 20),
+                              steps=1, warmup=0,
+                              direction=args.direction, mr=cand,
+                              device_index=dev_idx, timeout=120)
+            mr = cand
+            break
+        except RuntimeError as e:
+            probe_err = str(e)
+            print(f"# verbs mr={cand} unavailable: {probe_err}",
+                  flush=True)
+    if mr is None:
+        raise RuntimeError(
+            f"no usable verbs MR mode (last error: {probe_err})")
 
     if dist is not None:
         dist.barrier()
