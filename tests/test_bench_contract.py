@@ -202,3 +202,26 @@ def test_eight_rank_gloo_aggregate():
     # x 16 KiB over the max elapsed -> global_batch = msgs/step x 8
     assert r["config"]["global_batch"] == 16 * 8
     assert r["value"] > 0
+
+
+@pytest.mark.timeout(300)
+def test_verbs_read_direction():
+    """bench --direction read routes through the native verbs plane
+    (RDMA READ WRs) and keeps the JSON contract."""
+    if not os.path.exists(FAKEVERBS_BIN):
+        pytest.skip("fakeverbs harness not built")
+    env = dict(os.environ)
+    env["PYTHONPATH"] = ROOT + os.pathsep + env.get("PYTHONPATH", "")
+    env["ROCNR_FORCE_VERBS"] = "1"
+    env["ROCNR_VERBS_HARNESS"] = FAKEVERBS_BIN
+    out = subprocess.run(
+        [sys.executable, BENCH, "--gpus", "1", "--steps", "2",
+         "--warmup", "1", "--msg-bytes", "1048576",
+         "--region-bytes", "16777216", "--direction", "read"],
+        capture_output=True, text=True, timeout=240, env=env, cwd=ROOT)
+    assert out.returncode == 0, out.stderr
+    line = [l for l in out.stdout.splitlines() if l.startswith("{")][-1]
+    r = json.loads(line)
+    assert r["config"]["transport"] == "verbs"
+    assert r["config"]["direction"] == "read"
+    assert r["config"]["integrity"] == "ok"
