@@ -591,6 +591,11 @@ int ibv_post_send(struct ibv_qp* qp, struct ibv_send_wr* wr,
       FAIL("post_send: local MR lacks LOCAL_WRITE");
     FakeMr* remote = find_mr_by_key(wr->wr.rdma.rkey, true);
     if (remote) {
+#ifdef ROCNR_FAKEVERBS_PEER
+      if (remote->peer_handle && rocnr_glue_mr_dead(remote->peer_handle))
+        FAIL("post_send: peer MR invalidated (producer freed the memory "
+             "under it) — remote access error");
+#endif
       if (!range_ok(remote, wr->wr.rdma.remote_addr, wr->sg_list[0].length))
         FAIL("post_send: remote range out of MR bounds");
       uint8_t* rptr = mr_data_ptr(remote, wr->wr.rdma.remote_addr);

@@ -25,6 +25,10 @@
 
 #include "rocp2p_transport.h"
 
+#ifdef ROCNR_FAKEVERBS_PEER
+#include "../../module/shim/peer_glue.h"
+#endif
+
 namespace rocp2p {
 std::unique_ptr<Transport> make_fake_transport(const TransportConfig&);
 std::unique_ptr<Transport> make_hip_transport(const TransportConfig&);
@@ -169,6 +173,45 @@ int main(int argc, char** argv) {
     else if (a == "--serve") serve_port = atoi(next().c_str());
     else if (a == "--connect") connect_to = next();
     else if (a == "--remote-selftest") remote_selftest = true;
+    else if (a == "--peer-revoke-selftest") {
+#ifdef ROCNR_FAKEVERBS_PEER
+      // SURVEY §3.4 through the verbs surface: register a peer MR via
+      // the real bridge, move data, then the producer frees the
+      // memory under the live MR — the invalidation must tear the MR
+      // down (bridge free_cb -> IB-core invalidate) and subsequent
+      // posts must fail as remote-access errors, with deregistration
+      // still idempotent.
+      try {
+        cfg.verbs_mr = "peer";
+        cfg.msg_bytes = 64 << 10;
+        cfg.region_bytes = 8 << 20;
+        auto tp = make_verbs_transport(cfg);
+        tp->post_many(0, 16);
+        tp->flush();
+        rocnr_glue_revoke_last();  // GPU frees the region now
+        bool failed = false;
+        try {
+          tp->post_many(16, 1);
+          tp->flush();
+        } catch (const std::exception&) {
+          failed = true;
+        }
+        if (!failed) {
+          fprintf(stderr, "post after revoke unexpectedly succeeded\n");
+          return 1;
+        }
+        tp.reset();  // dereg path after invalidation: must be a no-op
+        printf("{\"mode\":\"peer-revoke-selftest\",\"result\":\"ok\"}\n");
+        return 0;
+      } catch (const std::exception& e) {
+        fprintf(stderr, "peer-revoke-selftest: %s\n", e.what());
+        return 3;
+      }
+#else
+      fprintf(stderr, "--peer-revoke-selftest needs the fullstack build\n");
+      return 2;
+#endif
+    }
     else if (a == "--lat") lat_iters = atoi(next().c_str());
     else if (a == "--steps") steps = atoi(next().c_str());
     else if (a == "--warmup") step_warmup = atoi(next().c_str());

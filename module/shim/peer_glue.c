@@ -31,9 +31,22 @@ int rocnr_glue_init(void)
 	return 0;
 }
 
+static uint64_t glue_last_va;
+
 uint64_t rocnr_glue_alloc(uint64_t bytes)
 {
-	return fake_kfd_alloc_backed(bytes);
+	uint64_t va = fake_kfd_alloc_backed(bytes);
+
+	if (va)
+		glue_last_va = va;
+	return va;
+}
+
+void rocnr_glue_revoke_last(void)
+{
+	if (glue_last_va)
+		fake_kfd_free(glue_last_va);
+	glue_last_va = 0;
 }
 
 void *rocnr_glue_vram_ptr(uint64_t va)
@@ -78,6 +91,13 @@ int rocnr_glue_reg_mr(uint64_t va, size_t size, void **handle_out,
 	*nsegs_inout = mr->nmap;
 	*handle_out = mr;
 	return 0;
+}
+
+int rocnr_glue_mr_dead(void *handle)
+{
+	struct fake_ib_mr *mr = handle;
+
+	return mr ? mr->dead : 1;
 }
 
 int rocnr_glue_dereg_mr(void *handle)

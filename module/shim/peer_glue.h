@@ -58,6 +58,17 @@ int rocnr_glue_reg_mr(uint64_t va, size_t size, void **handle_out,
  * through the real bridge. */
 int rocnr_glue_dereg_mr(void *handle);
 
+/* Has this MR been torn down by the invalidation path (producer freed
+ * the memory under it)?  The verbs layer checks this before touching
+ * the MR's (now reclaimed) pages — modeling the remote-access errors
+ * a real HCA raises after invalidation. */
+int rocnr_glue_mr_dead(void *handle);
+
+/* Revoke the most recent glue allocation (GPU-frees it under any live
+ * MRs): drives free_callback -> bridge -> IB-core invalidate.  For the
+ * harness revoke selftest. */
+void rocnr_glue_revoke_last(void);
+
 /* Bus address -> CPU pointer for the fake NIC's data plane (NULL if
  * the bus range is not backed). */
 void *rocnr_glue_bus_ptr(uint64_t bus);

@@ -298,3 +298,16 @@ def test_two_process_fabric_peer_mr(built_fullstack, tmp_path):
     deployment shape (ib_write_bw server with GPU memory) across
     genuine process boundaries."""
     _two_process_fabric(built_fullstack, "peer", tmp_path)
+
+
+@pytest.mark.timeout(300)
+def test_peer_revoke_through_verbs_surface(built_fullstack):
+    """SURVEY §3.4 visible from L5: the producer frees memory under a
+    live bridge-registered MR; the invalidation (bridge free_cb ->
+    IB-core invalidate -> teardown) must make subsequent posts fail as
+    remote-access errors and leave deregistration idempotent."""
+    out = subprocess.run([built_fullstack, "--peer-revoke-selftest"],
+                         capture_output=True, text=True, timeout=120)
+    assert out.returncode == 0, out.stdout + out.stderr
+    assert '"result":"ok"' in out.stdout
+    assert "peer MR invalidated" in out.stderr
