@@ -59,6 +59,31 @@ def main():
         print(f"config matrix       : {len(done)} configs measured "
               f"({', '.join(done)})")
 
+    # --- round 2 ---
+    b = load("r2_bench_64g_region.json")
+    if b:
+        print(f"r2 64 GiB region    : {b['value']} GB/s "
+              f"(integrity {b['config']['integrity']})")
+    b = load("bench_sustained_r2.json")
+    if b:
+        tb = b['steps'] * 1.074 / 1000  # 1 GiB region per step
+        print(f"r2 5-min sustained  : {b['value']} GB/s "
+              f"({b['steps']} steps, ~{tb:.1f} TB)")
+    b = load("bench_verbs_2rank.json")
+    if b:
+        print(f"r2 bench verbs path : {b['value']} GB/s aggregate, "
+              f"{b['config']['parallelism']}, mr="
+              f"{b['config'].get('verbs_mr')} (real VRAM dmabuf)")
+    for name, label in (("dmabuf_write.json", "r2 dmabuf BAR write"),
+                        ("dmabuf_16g.json", "r2 dmabuf 16 GiB")):
+        d = load(name)
+        if d:
+            print(f"{label:<20}: {d['gbps']} GB/s "
+                  f"(integrity {d['integrity']})")
+    for f in sorted(glob.glob(os.path.join(P, "r2_soak_*.txt"))):
+        with open(f) as fh:
+            print(os.path.basename(f), "->", fh.read().strip()[-110:])
+
 
 if __name__ == "__main__":
     main()
