@@ -311,3 +311,12 @@ def test_peer_revoke_through_verbs_surface(built_fullstack):
     assert out.returncode == 0, out.stdout + out.stderr
     assert '"result":"ok"' in out.stdout
     assert "peer MR invalidated" in out.stderr
+
+
+@pytest.mark.timeout(300)
+def test_two_process_fabric_large_messages(built_fakeverbs, tmp_path):
+    """Messages larger than the fabric's 1 MiB slot payload must be
+    chunked transparently (one WR = several fabric ops) with exact
+    remote delivery."""
+    _two_process_fabric(built_fakeverbs, "host", tmp_path,
+                        msg="4194304", region="16777216")
