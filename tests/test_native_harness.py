@@ -320,3 +320,21 @@ def test_two_process_fabric_large_messages(built_fakeverbs, tmp_path):
     remote delivery."""
     _two_process_fabric(built_fakeverbs, "host", tmp_path,
                         msg="4194304", region="16777216")
+
+
+@pytest.mark.parametrize("n", ["2", "4"])
+@pytest.mark.timeout(300)
+def test_verbs_multi_qp_fanout(built_fakeverbs, n):
+    """--gpus N over the verbs backend: N concurrent workers, each
+    with its own QP pair/MRs/CQ in one process (the 8-GPU node fan-out
+    shape at the verbs layer); integrity per worker, aggregate
+    reported."""
+    out = subprocess.run(
+        [built_fakeverbs, "--transport", "verbs", "--gpus", n,
+         "--msg", "65536", "--region", "4194304", "--secs", "0.1",
+         "--json"],
+        capture_output=True, text=True, timeout=120)
+    assert out.returncode == 0, out.stdout + out.stderr
+    r = json.loads(out.stdout.strip().splitlines()[-1])
+    assert r["gpus"] == int(n)
+    assert r["integrity"] == "ok"
