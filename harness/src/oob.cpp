@@ -7,6 +7,7 @@
 #include <sys/socket.h>
 #include <unistd.h>
 
+#include <cstdlib>
 #include <cstring>
 #include <sstream>
 #include <stdexcept>
@@ -75,7 +76,14 @@ OobServer::OobServer(int port) {
   setsockopt(lfd_, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
   sockaddr_in a{};
   a.sin_family = AF_INET;
-  a.sin_addr.s_addr = htonl(INADDR_LOOPBACK);
+  // Listen on all interfaces so the documented two-host client/server
+  // mode actually accepts remote clients (ib_write_bw server shape);
+  // ROCP2P_OOB_BIND=127.0.0.1 restricts to loopback.
+  a.sin_addr.s_addr = htonl(INADDR_ANY);
+  if (const char* b = getenv("ROCP2P_OOB_BIND")) {
+    if (inet_pton(AF_INET, b, &a.sin_addr) != 1)
+      throw std::runtime_error("ROCP2P_OOB_BIND: bad IPv4 address");
+  }
   a.sin_port = htons((uint16_t)port);
   if (bind(lfd_, (sockaddr*)&a, sizeof(a)) != 0)
     throw std::runtime_error("oob bind failed");
