@@ -112,3 +112,16 @@ def test_abi_probe_classifies(tmp_path, peer, amdr, u64, dmadev):
     assert out.returncode == 0, out.stdout + out.stderr
     assert f"ROCNR_PEER_MEM_CORE_CONTEXT_U64={u64}" in out.stdout
     assert f"ROCNR_AMD_RDMA_HAS_DMA_DEV={dmadev}" in out.stdout
+
+
+def test_first_hca_bringup_dry_run():
+    """The executable bring-up sequence (RUNBOOK as a script) must
+    dry-run cleanly: artifact checks pass and every step prints."""
+    import subprocess
+    out = subprocess.run(
+        ["bash", os.path.join(ROOT, "tools", "first_hca_bringup.sh"),
+         "--dry-run"],
+        capture_output=True, text=True, timeout=60)
+    assert out.returncode == 0, out.stdout + out.stderr
+    assert "DONE" in out.stdout
+    assert "chain sweep" in out.stdout
